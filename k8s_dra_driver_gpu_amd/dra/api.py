@@ -1,11 +1,20 @@
 """DRA kubelet-plugin gRPC message types + service plumbing.
 
-Wire-compatible with ``k8s.io/kubelet/pkg/apis/dra/v1beta1`` (service
-``v1beta1.DRAPlugin``; also served under ``v1.DRAPlugin``) and
-``pluginregistration/v1`` (service ``pluginregistration.Registration``) —
-the same contract the reference consumes through the
-``k8s.io/dynamic-resource-allocation/kubeletplugin`` helper
-(ref gpu-kubelet-plugin/driver.go:131-149).
+Wire-compatible with ``k8s.io/kubelet/pkg/apis/dra/v1beta1``: the gRPC
+service is registered under its fully-qualified proto name
+``k8s.io.kubelet.pkg.apis.dra.v1beta1.DRAPlugin`` (and the v1 surface under
+``k8s.io.kubelet.pkg.apis.dra.v1.DRAPlugin``), exactly as the kubelet dials
+it (ref vendor k8s.io/kubelet/pkg/apis/dra/v1beta1/api.proto:21 ``package
+k8s.io.kubelet.pkg.apis.dra.v1beta1``; api_grpc.pb.go:55-56,190
+``ServiceName: "k8s.io.kubelet.pkg.apis.dra.v1beta1.DRAPlugin"``).
+
+Plugin registration speaks ``pluginregistration/v1`` (service
+``pluginregistration.Registration``) and advertises the DRA *service
+identifiers* ``v1.DRAPlugin`` / ``v1beta1.DRAPlugin`` as its
+"supported versions" — NOT bare API versions (ref vendor
+k8s.io/kubelet/pkg/apis/dra/v1beta1/types.go:23 ``DRAPluginService =
+"v1beta1.DRAPlugin"``; dynamic-resource-allocation/kubeletplugin/
+draplugin.go:755-763 appends v1 then v1beta1).
 """
 
 from __future__ import annotations
@@ -86,6 +95,22 @@ class RegistrationStatusResponse(Message):
 
 DRA_PLUGIN_TYPE = "DRAPlugin"  # kubelet plugin registration type
 
+# Fully-qualified gRPC service names (proto package + service). The kubelet's
+# client stubs invoke e.g.
+# /k8s.io.kubelet.pkg.apis.dra.v1beta1.DRAPlugin/NodePrepareResources
+# (ref vendor .../dra/v1beta1/api_grpc.pb.go:55-56,190; v1: api_grpc.pb.go:190).
+DRA_SERVICE_FULL = {
+    "v1": "k8s.io.kubelet.pkg.apis.dra.v1.DRAPlugin",
+    "v1beta1": "k8s.io.kubelet.pkg.apis.dra.v1beta1.DRAPlugin",
+}
+
+# Service *identifiers* advertised through pluginregistration GetInfo's
+# supported_versions field — the kubelet matches these strings to decide which
+# DRA service variant to dial (ref vendor .../dra/{v1,v1beta1}/types.go:23,
+# kubeletplugin/noderegistrar.go:39). Order matches the reference helper:
+# v1 first, then v1beta1 (draplugin.go:755-763).
+DRA_SERVICE_IDENTIFIERS = ("v1.DRAPlugin", "v1beta1.DRAPlugin")
+
 
 def _unary(handler: Callable[[Message], Message], req_cls):
     def call(request_bytes, context):
@@ -119,7 +144,7 @@ class DRAPluginServicer:
         }
         for pkg in packages:
             server.add_generic_rpc_handlers(
-                (grpc.method_handlers_generic_handler(f"{pkg}.DRAPlugin", handlers),)
+                (grpc.method_handlers_generic_handler(DRA_SERVICE_FULL[pkg], handlers),)
             )
 
 
@@ -127,7 +152,7 @@ class RegistrationServicer:
     """kubelet pluginregistration/v1 service (ref: kubeletplugin helper
     registers via the plugins_registry socket)."""
 
-    def __init__(self, name: str, endpoint: str, supported_versions=("v1beta1", "v1")):
+    def __init__(self, name: str, endpoint: str, supported_versions=DRA_SERVICE_IDENTIFIERS):
         self.name = name
         self.endpoint = endpoint
         self.supported_versions = list(supported_versions)
@@ -165,7 +190,7 @@ class DRAPluginClient:
 
     def __init__(self, target: str, package: str = "v1beta1", channel: Optional[grpc.Channel] = None):
         self.channel = channel or grpc.insecure_channel(target)
-        p = f"/{package}.DRAPlugin"
+        p = f"/{DRA_SERVICE_FULL[package]}"
         self._prepare = self.channel.unary_unary(
             f"{p}/NodePrepareResources",
             request_serializer=lambda m: m.to_bytes(),

@@ -163,7 +163,9 @@ class TestPluginService:
         assert info.type == "DRAPlugin"
         assert info.name == "gpu.amd.com"
         assert info.endpoint == socks["dra"]
-        assert "v1beta1" in info.supported_versions
+        # The kubelet matches DRA *service identifiers*, not bare versions
+        # (ref vendor .../dra/v1beta1/types.go:23).
+        assert info.supported_versions == ["v1.DRAPlugin", "v1beta1.DRAPlugin"]
         reg.notify(True)
         assert driver.registration.registered is True
         reg.close()
