@@ -69,6 +69,11 @@ class PreparedDevice:
     compute_mode: str = api_field("computeMode", default="")
     memory_mode: str = api_field("memoryMode", default="")
     partition_index: int = api_field("partitionIndex", default=0)
+    # PCI bus ID, checkpointed for type=vfio: a GPU bound to vfio-pci has no
+    # drm card so UUID lookup fails after a plugin restart — unbind must use
+    # the address directly (ref prepared.go checkpoints PciBusID for Vfio).
+    # Optional so existing gpu/partition payloads keep identical bytes.
+    pci_bus_id: Optional[str] = api_field("pciBusID", default=None)
     cdi_device_ids: List[str] = api_field("cdiDeviceIDs", default_factory=list)
     device_nodes: List[str] = api_field("deviceNodes", default_factory=list)
     config: Optional[Dict[str, Any]] = api_field("config", default=None)  # opaque config used

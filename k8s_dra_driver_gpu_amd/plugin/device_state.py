@@ -234,6 +234,7 @@ class DeviceState:
             type="vfio",
             name=alloc.device,
             uuid=gpu.uuid,
+            pci_bus_id=gpu.pci_bus_id,
             device_nodes=[n.path for n in edits.device_nodes],
             config=serde.to_dict(cfg),
         )
@@ -408,12 +409,28 @@ class DeviceState:
 
     def _undo_device(self, d: PreparedDevice, checkpoint=None, skip_claim: str = "") -> None:
         if d.type == "vfio" and self.vfio is not None:
-            gpu = self.devicelib.gpu_by_uuid(d.uuid)
-            if gpu is not None:
-                try:
-                    self.vfio.unprepare(gpu.pci_bus_id)
-                except Exception:
-                    logger.exception("vfio unbind of %s failed", d.name)
+            # Resolve via the checkpointed PCI address first: a GPU bound to
+            # vfio-pci has no drm card, so gpu_by_uuid (which walks
+            # /sys/class/drm) cannot see it after a plugin restart. UUID
+            # lookup is only a fallback for pre-upgrade checkpoints.
+            pci = d.pci_bus_id
+            if not pci:
+                gpu = self.devicelib.gpu_by_uuid(d.uuid)
+                pci = gpu.pci_bus_id if gpu is not None else None
+            if pci is None:
+                logger.error(
+                    "vfio unprepare of %s: no checkpointed PCI bus ID and "
+                    "UUID %s not resolvable (GPU likely still bound to "
+                    "vfio-pci) — device is STRANDED on vfio-pci; rebind to "
+                    "amdgpu manually or re-prepare",
+                    d.name,
+                    d.uuid,
+                )
+                return
+            try:
+                self.vfio.unprepare(pci)
+            except Exception:
+                logger.exception("vfio unbind of %s (%s) failed", d.name, pci)
             return
         if d.type == "partition" and d.parent_uuid:
             # Return parent to SPX only when no OTHER claim still holds a

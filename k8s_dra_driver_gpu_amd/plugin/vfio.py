@@ -72,6 +72,27 @@ class VfioPciManager:
     def iommufd_available(self) -> bool:
         return os.path.exists(os.path.join(self.dev_root, "iommu"))
 
+    def iommufd_cdev_name(self, pci: str) -> str:
+        """Discover the per-device IOMMUFD cdev name (``vfioX``).
+
+        The cdev index is per-device and UNRELATED to the IOMMU group
+        number; the kernel publishes it as
+        ``/sys/bus/pci/devices/<addr>/vfio-dev/vfio<X>`` once the device is
+        bound to vfio-pci (ref vfio-device.go reads this dir and errors if
+        missing). Raises VfioError if absent.
+        """
+        d = os.path.join(self.pci_dev_dir(pci), "vfio-dev")
+        try:
+            names = sorted(n for n in os.listdir(d) if n.startswith("vfio"))
+        except OSError:
+            names = []
+        if not names:
+            raise VfioError(
+                f"IOMMUFD selected but {d} has no vfio cdev entry — "
+                f"is {pci} bound to vfio-pci with VFIO_DEVICE_CDEV enabled?"
+            )
+        return names[0]
+
     # -- busy wait ----------------------------------------------------------
 
     def _default_busy_check(self, pci: str) -> bool:
@@ -138,7 +159,9 @@ class VfioPciManager:
             and self.iommufd_available()
         )
         if use_iommufd:
-            vfio_path = os.path.join(self.dev_root, "vfio", "devices", f"vfio{group}")
+            # cdev index is per-device, not the IOMMU group number
+            cdev = self.iommufd_cdev_name(pci)
+            vfio_path = os.path.join(self.dev_root, "vfio", "devices", cdev)
         else:
             vfio_path = os.path.join(self.dev_root, "vfio", group)
         return VfioDeviceInfo(

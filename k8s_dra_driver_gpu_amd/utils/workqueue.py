@@ -117,6 +117,11 @@ class WorkQueue:
         self._shutdown = False
         self._idle_cond = threading.Condition(self._lock)
         self._inflight = 0
+        # Keys currently executing. With workers>1 a re-enqueued key must NOT
+        # run concurrently with its still-in-flight predecessor — client-go's
+        # workqueue guarantees per-key serialization via its dirty/processing
+        # sets; we defer dispatch of a running key until the run finishes.
+        self._running: set = set()
         self._threads = [
             threading.Thread(target=self._run, name=f"{name}-{i}", daemon=True)
             for i in range(workers)
@@ -139,6 +144,7 @@ class WorkQueue:
     def _run(self) -> None:
         while True:
             with self._lock:
+                item = None
                 while True:
                     if self._shutdown:
                         return
@@ -146,12 +152,31 @@ class WorkQueue:
                     # Drop superseded entries lazily.
                     while self._heap and self._latest.get(self._heap[0].key) != self._heap[0].seq:
                         heapq.heappop(self._heap)
-                    if self._heap and self._heap[0].ready_at <= now:
-                        item = heapq.heappop(self._heap)
+                    # Earliest ready item whose key is not in flight; ready
+                    # items for running keys are deferred (per-key
+                    # serialization) and re-pushed.
+                    deferred = []
+                    while self._heap and self._heap[0].ready_at <= now:
+                        cand = heapq.heappop(self._heap)
+                        if self._latest.get(cand.key) != cand.seq:
+                            continue  # superseded
+                        if cand.key in self._running:
+                            deferred.append(cand)
+                            continue
+                        item = cand
+                        break
+                    for b in deferred:
+                        heapq.heappush(self._heap, b)
+                    if item is not None:
                         self._latest.pop(item.key, None)
+                        self._running.add(item.key)
                         self._inflight += 1
                         break
                     timeout = (self._heap[0].ready_at - now) if self._heap else None
+                    if timeout is not None and timeout <= 0:
+                        # Head is ready but its key is running: wait for the
+                        # completion notify, not a timed spin.
+                        timeout = None
                     self._cond.wait(timeout=timeout)
             try:
                 item.work()
@@ -169,7 +194,11 @@ class WorkQueue:
                         self._cond.notify()
             finally:
                 with self._lock:
+                    self._running.discard(item.key)
                     self._inflight -= 1
+                    # Wake workers: a deferred same-key item may now be
+                    # dispatchable.
+                    self._cond.notify_all()
                     self._idle_cond.notify_all()
 
     def wait_idle(self, timeout: float = 10.0) -> bool:

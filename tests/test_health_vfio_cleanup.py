@@ -280,6 +280,36 @@ class TestVfio:
         with pytest.raises(VfioError, match="SR-IOV"):
             mgr.prepare(gpu)
 
+    def _iommufd_cfg(self):
+        from k8s_dra_driver_gpu_amd.api.configs import (
+            IOMMU_PREFER_IOMMUFD,
+            IommuConfig,
+            VfioDeviceConfig,
+        )
+
+        return VfioDeviceConfig(
+            iommu=IommuConfig(backend_policy=IOMMU_PREFER_IOMMUFD, enable_api_device=True)
+        )
+
+    def test_iommufd_cdev_discovered_from_sysfs(self, vfio_env):
+        """The IOMMUFD cdev index is per-device, NOT the IOMMU group number:
+        it must be read from <pci>/vfio-dev/vfioX (ref vfio-device.go)."""
+        vt, mgr, gpu = vfio_env
+        open(os.path.join(vt.dev, "iommu"), "w").close()
+        devdir = os.path.join(vt.sysfs, "bus", "pci", "devices", vt.pci)
+        os.makedirs(os.path.join(devdir, "vfio-dev", "vfio7"))
+        info = mgr.prepare(gpu, self._iommufd_cfg())
+        assert info.vfio_dev_path.endswith("/vfio/devices/vfio7")
+        assert "42" not in os.path.basename(info.vfio_dev_path)
+
+    def test_iommufd_missing_cdev_fails_prepare(self, vfio_env):
+        """IOMMUFD selected but no vfio-dev entry: prepare must fail loudly
+        rather than inject a nonexistent node."""
+        vt, mgr, gpu = vfio_env
+        open(os.path.join(vt.dev, "iommu"), "w").close()
+        with pytest.raises(VfioError, match="vfio cdev"):
+            mgr.prepare(gpu, self._iommufd_cfg())
+
 
 class TestVfioInDeviceState:
     """VFIO wired through the Prepare/Unprepare state machine."""
@@ -313,6 +343,24 @@ class TestVfioInDeviceState:
         assert mgr.current_driver(vt.pci) == "vfio-pci"
         pc = ds.checkpoints.load().get_claim(UID1)
         assert pc.devices[0].type == "vfio"
+        assert pc.devices[0].pci_bus_id == vt.pci  # checkpointed for restart
+        ds.unprepare(UID1)
+        assert mgr.current_driver(vt.pci) == "amdgpu"
+
+    def test_vfio_unprepare_after_restart_uses_checkpointed_pci(self, vfio_state):
+        """After a plugin restart the GPU is bound to vfio-pci and has no drm
+        card, so UUID lookup fails — unprepare must fall back to the
+        checkpointed PCI bus ID instead of silently stranding the device."""
+        vt, mgr, ds = vfio_state
+        ds.prepare(
+            AllocatedClaim(
+                ref=ClaimRef(namespace="d", name="v", uid=UID1),
+                devices=[AllocatedDevice(device="gpu-0-vfio")],
+            )
+        )
+        assert mgr.current_driver(vt.pci) == "vfio-pci"
+        # simulate restart: devicelib can no longer resolve the UUID
+        ds.devicelib.gpu_by_uuid = lambda uuid: None
         ds.unprepare(UID1)
         assert mgr.current_driver(vt.pci) == "amdgpu"
 

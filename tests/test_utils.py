@@ -169,6 +169,61 @@ class TestWorkQueue:
         assert log[-1] == "ok"
         assert log.count("ok") == 1
 
+    def test_per_key_serialization_with_many_workers(self):
+        """A re-enqueued key must never run concurrently with its in-flight
+        predecessor (client-go dirty/processing semantics); other keys keep
+        running in parallel."""
+        wq = WorkQueue(name="t5", workers=4)
+        lock = threading.Lock()
+        running = {"k": 0}
+        max_concurrent = [0]
+        first_started = threading.Event()
+        release_first = threading.Event()
+        other_done = threading.Event()
+
+        def work_k(block):
+            with lock:
+                running["k"] += 1
+                max_concurrent[0] = max(max_concurrent[0], running["k"])
+            first_started.set()
+            if block:
+                release_first.wait(2.0)
+            with lock:
+                running["k"] -= 1
+
+        wq.enqueue("k", lambda: work_k(True))
+        assert first_started.wait(2.0)
+        # queued while the predecessor is still executing
+        wq.enqueue("k", lambda: work_k(False))
+        # an unrelated key is NOT blocked by k's serialization
+        wq.enqueue("other", other_done.set)
+        assert other_done.wait(2.0)
+        release_first.set()
+        assert wq.wait_idle(5.0)
+        wq.shutdown()
+        assert max_concurrent[0] == 1
+
+    def test_deferred_key_runs_after_predecessor(self):
+        """The deferred enqueue is not lost: it runs once the in-flight run
+        finishes."""
+        wq = WorkQueue(name="t6", workers=2)
+        order = []
+        gate = threading.Event()
+        started = threading.Event()
+
+        def first():
+            started.set()
+            gate.wait(2.0)
+            order.append("first")
+
+        wq.enqueue("k", first)
+        assert started.wait(2.0)
+        wq.enqueue("k", lambda: order.append("second"))
+        gate.set()
+        assert wq.wait_idle(5.0)
+        wq.shutdown()
+        assert order == ["first", "second"]
+
 
 class TestFeatureGates:
     def test_defaults(self):
