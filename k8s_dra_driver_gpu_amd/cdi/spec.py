@@ -112,6 +112,9 @@ class CdiHandler:
     def common_spec_path(self) -> str:
         return os.path.join(self.cdi_root, f"{self.driver_name}-common.json")
 
+    def standard_spec_path(self) -> str:
+        return os.path.join(self.cdi_root, f"{self.driver_name}-standard.json")
+
     # -- device-node edits ---------------------------------------------------
 
     def _host(self, path: str) -> str:
@@ -206,6 +209,36 @@ class CdiHandler:
             ],
         }
         path = self.common_spec_path()
+        tmp = path + ".tmp"
+        with open(tmp, "w", encoding="utf-8") as f:
+            json.dump(spec, f, indent=2, sort_keys=True)
+        os.replace(tmp, path)
+        return path
+
+    def write_standard_spec(self, render_minors: Optional[List[int]] = None) -> str:
+        """Boot-time "standard" management spec: one device named ``all``
+        carrying the full driver injection (/dev/kfd, every render node, ROCm
+        runtime mounts). Daemon-type prepared devices reference
+        ``<vendor>/<klass>=all`` IN ADDITION to their per-claim device, so the
+        fabric-daemon container gets driver enablement even though the
+        per-claim spec carries only config-state edits (ref
+        compute-domain-kubelet-plugin/cdi.go:142-206 CreateStandardDeviceSpecFile,
+        device_state.go:467 GetStandardDevice)."""
+        edits = ContainerEdits()
+        edits.device_nodes.append(self.kfd_node())
+        for rm in render_minors or []:
+            edits.device_nodes.append(self.render_node(rm))
+        edits.env.append(f"AMDDRA_DRIVER={self.driver_name}")
+        # keep lower-level runtimes from double-injecting devices
+        # (the reference sets NVIDIA_VISIBLE_DEVICES=void for the same reason)
+        edits.env.append("AMD_VISIBLE_DEVICES=void")
+        spec = {
+            "cdiVersion": CDI_VERSION,
+            "kind": f"{self.vendor}/{self.klass}",
+            "devices": [CdiDevice(name="all", edits=edits).render()],
+        }
+        spec["devices"][0]["containerEdits"]["mounts"] = self._rocm_runtime_mounts()
+        path = self.standard_spec_path()
         tmp = path + ".tmp"
         with open(tmp, "w", encoding="utf-8") as f:
             json.dump(spec, f, indent=2, sort_keys=True)

@@ -44,6 +44,7 @@ from ..dra import api as dra
 from ..k8s.client import Client
 from ..plugin.checkpoint import (
     PREPARE_COMPLETED,
+    PREPARE_STARTED,
     CheckpointManager,
     ClaimRef,
     PreparedClaim,
@@ -97,6 +98,15 @@ class ComputeDomainPlugin(dra.DRAPluginServicer):
         # fallback): with strict_fabric, a GPU node without a derivable xGMI
         # clique is a hard error instead of clique-less operation.
         self.strict_fabric = strict_fabric
+        # Boot-time "standard" management spec: full driver injection under
+        # device name `all`; per-claim specs carry only config-state edits
+        # (ref cdi.go:142-206). Written once per plugin start.
+        try:
+            self.cdi.write_standard_spec(
+                sorted(g.render_minor for g in self.devicelib.gpus())
+            )
+        except Exception:
+            logger.exception("failed to write boot-time standard CDI spec")
 
     # ------------------------------------------------------------------
     # Device model: daemon-0 + channel-0
@@ -205,6 +215,19 @@ class ComputeDomainPlugin(dra.DRAPluginServicer):
         return claim, results, configs
 
     def _prepare_claim(self, claim_msg) -> dra.NodePrepareResourceResponse:
+        """Two-phase prepare (ref device_state.go:186-256):
+
+        1. durable ``PrepareStarted`` intent BEFORE any side effect, with
+           channel exclusivity asserted atomically inside the RMW;
+        2. side effects (domain dir, node label, readiness gate, CDI spec);
+        3. ``PrepareCompleted`` commit, exclusivity re-asserted atomically.
+
+        A crash between 1 and 3 leaves a ``PrepareStarted`` entry that
+        annotates the potentially-partial prepare: a retry re-enters and
+        finishes it, and the async cleanup manager reaps it if the claim
+        goes stale (ref device_state.go:239-243, cleanup.go:117-125).
+        A PermanentError after phase 1 (e.g. losing a channel race) rolls
+        back this claim's side effects."""
         uid = claim_msg.uid
         existing = self.checkpoints.load().get_claim(uid)
         if existing is not None and existing.state == PREPARE_COMPLETED:
@@ -219,57 +242,142 @@ class ComputeDomainPlugin(dra.DRAPluginServicer):
                 ]
             )
         _, results, configs = self._resolve(claim_msg.namespace, claim_msg.name, uid)
-        devices: List[PreparedDevice] = []
-        cdi_devices: List[CdiDevice] = []
-        out_devices: List[dra.Device] = []
+        intents = []  # (device_name, cfg)
         for res in results:
             if res.get("driver") not in (None, self.driver_name):
                 continue
             device_name = res.get("device", "")
             cfg = self._config_for(res, configs)
-            if isinstance(cfg, ComputeDomainDaemonConfig):
-                pd, cd_dev = self._prepare_daemon(claim_msg, device_name, cfg)
-            elif isinstance(cfg, ComputeDomainChannelConfig):
-                pd, cd_dev = self._prepare_channel(claim_msg, device_name, cfg)
-            else:
+            if not isinstance(cfg, (ComputeDomainDaemonConfig, ComputeDomainChannelConfig)):
                 raise PermanentError(
                     f"device {device_name} has no ComputeDomain opaque config"
                 )
-            devices.append(pd)
-            cdi_devices.append(cd_dev)
-        cdi_ids = self.cdi.write_claim_spec(uid, cdi_devices)
-        for pd, cid in zip(devices, cdi_ids):
-            pd.cdi_device_ids = [cid]
-            out_devices.append(
-                dra.Device(
-                    pool_name=self.node_name, device_name=pd.name, cdi_device_ids=[cid]
-                )
-            )
+            intents.append((device_name, cfg))
         ref = ClaimRef(namespace=claim_msg.namespace, name=claim_msg.name, uid=uid)
 
-        def commit(data):
-            # channel exclusivity re-checked ATOMICALLY inside the RMW: two
-            # concurrent prepares must not both take channel 0 of a domain
-            # (ref assertImexChannelNotAllocated, device_state.go:729-757)
-            for pd in devices:
-                if pd.type != "channel":
-                    continue
-                for ouid, pc in data.claims().items():
-                    if ouid == uid or pc is None:
-                        continue
-                    for od in pc.devices or []:
-                        if (od.type == "channel" and od.uuid == pd.uuid
-                                and od.partition_index == pd.partition_index):
-                            raise PermanentError(
-                                f"channel {pd.partition_index} of domain {pd.uuid} "
-                                f"already allocated to claim {ouid}"
-                            )
+        # Phase 1: PrepareStarted — durable intent with the would-be devices,
+        # channel reservation checked atomically in the same RMW.
+        def start(data):
+            intent_devices = []
+            for device_name, cfg in intents:
+                if isinstance(cfg, ComputeDomainChannelConfig):
+                    self._check_channel_free(data, uid, cfg.domain_id, 0)
+                    intent_devices.append(
+                        PreparedDevice(
+                            type="channel", name=device_name, uuid=cfg.domain_id
+                        )
+                    )
+                else:
+                    intent_devices.append(
+                        PreparedDevice(
+                            type="daemon", name=device_name, uuid=cfg.domain_id
+                        )
+                    )
             data.set_claim(
-                uid, PreparedClaim(state=PREPARE_COMPLETED, claim=ref, devices=devices)
+                uid,
+                PreparedClaim(state=PREPARE_STARTED, claim=ref, devices=intent_devices),
             )
 
-        self.checkpoints.update(commit)
-        return dra.NodePrepareResourceResponse(devices=out_devices)
+        self.checkpoints.update(start)
+
+        created_dirs: List[str] = []
+        try:
+            devices: List[PreparedDevice] = []
+            cdi_devices: List[CdiDevice] = []
+            out_devices: List[dra.Device] = []
+            for device_name, cfg in intents:
+                if isinstance(cfg, ComputeDomainDaemonConfig):
+                    pd, cd_dev = self._prepare_daemon(
+                        claim_msg, device_name, cfg, created_dirs
+                    )
+                else:
+                    pd, cd_dev = self._prepare_channel(
+                        claim_msg, device_name, cfg, created_dirs
+                    )
+                devices.append(pd)
+                cdi_devices.append(cd_dev)
+            cdi_ids = self.cdi.write_claim_spec(uid, cdi_devices)
+            for pd, cid in zip(devices, cdi_ids):
+                ids = [cid]
+                if pd.type == "daemon":
+                    # daemon containers also get the boot-time standard
+                    # (management) device for full driver injection
+                    # (ref device_state.go:467 GetStandardDevice — empty for
+                    # channel-type devices)
+                    ids = [self.cdi.qualified_name("all"), cid]
+                pd.cdi_device_ids = ids
+                out_devices.append(
+                    dra.Device(
+                        pool_name=self.node_name, device_name=pd.name, cdi_device_ids=ids
+                    )
+                )
+
+            # Phase 2: PrepareCompleted — exclusivity re-asserted atomically
+            # inside the commit RMW (multi-process safe, strictly stronger
+            # than the reference's in-process lock;
+            # ref assertImexChannelNotAllocated device_state.go:729-757).
+            def commit(data):
+                for pd in devices:
+                    if pd.type == "channel":
+                        self._check_channel_free(
+                            data, uid, pd.uuid, pd.partition_index
+                        )
+                data.set_claim(
+                    uid,
+                    PreparedClaim(state=PREPARE_COMPLETED, claim=ref, devices=devices),
+                )
+
+            self.checkpoints.update(commit)
+            return dra.NodePrepareResourceResponse(devices=out_devices)
+        except PermanentError:
+            self._rollback_started(uid, created_dirs)
+            raise
+        # TransientErrors propagate with the PrepareStarted entry left in
+        # place (deliberate): the retry loop re-enters, and stale entries are
+        # reaped by cleanup_stale_claims.
+
+    def _check_channel_free(
+        self, data, claim_uid: str, domain_id: str, channel: int = 0
+    ) -> None:
+        """Channel exclusivity against COMPLETED claims only: a claim stuck
+        in PrepareStarted is either about to retry (we win the race and it
+        re-checks) or stale (cleanup reaps it) — ref device_state.go:736-745."""
+        for ouid, pc in data.claims().items():
+            if ouid == claim_uid or pc is None or pc.state != PREPARE_COMPLETED:
+                continue
+            for od in pc.devices or []:
+                if (
+                    od.type == "channel"
+                    and od.uuid == domain_id
+                    and od.partition_index == channel
+                ):
+                    raise PermanentError(
+                        f"channel {channel} of domain {domain_id} "
+                        f"already allocated to claim {ouid}"
+                    )
+
+    def _rollback_started(self, uid: str, created_dirs: List[str]) -> None:
+        """Undo a failed prepare's side effects: claim CDI spec, checkpoint
+        entry, and any domain dir THIS prepare created that no surviving
+        claim references."""
+        try:
+            self.cdi.delete_claim_spec(uid)
+        except Exception:
+            logger.exception("rollback: CDI spec removal for %s failed", uid)
+        try:
+            self.checkpoints.update(lambda d: d.remove_claim(uid))
+        except Exception:
+            logger.exception("rollback: checkpoint removal for %s failed", uid)
+        if not created_dirs:
+            return
+        cp = self.checkpoints.load()
+        referenced = set()
+        for pc in cp.claims().values():
+            for d in (pc.devices if pc else []) or []:
+                referenced.add(d.uuid)
+        for ddir in created_dirs:
+            if os.path.basename(ddir) not in referenced:
+                shutil.rmtree(ddir, ignore_errors=True)
 
     def _config_for(self, result: Dict[str, Any], configs: List[Dict[str, Any]]):
         request = result.get("request", "")
@@ -293,12 +401,26 @@ class ComputeDomainPlugin(dra.DRAPluginServicer):
     def domain_dir(self, cd_uid: str) -> str:
         return os.path.join(self.domains_dir, cd_uid)
 
-    def _prepare_daemon(self, claim_msg, device_name: str, cfg: ComputeDomainDaemonConfig):
+    def _ensure_domain_dir(self, cd_uid: str, created_dirs: List[str]) -> str:
+        """Create the per-domain dir, recording it for rollback only when
+        THIS call created it (a pre-existing dir belongs to other claims)."""
+        ddir = self.domain_dir(cd_uid)
+        if not os.path.isdir(ddir):
+            created_dirs.append(ddir)
+        os.makedirs(os.path.join(ddir, "shared"), exist_ok=True)
+        return ddir
+
+    def _prepare_daemon(
+        self,
+        claim_msg,
+        device_name: str,
+        cfg: ComputeDomainDaemonConfig,
+        created_dirs: List[str],
+    ):
         cd = self._get_cd_by_uid(cfg.domain_id)
         if cd is None:
             raise TransientError(f"ComputeDomain {cfg.domain_id} not found")
-        ddir = self.domain_dir(cfg.domain_id)
-        os.makedirs(os.path.join(ddir, "shared"), exist_ok=True)
+        ddir = self._ensure_domain_dir(cfg.domain_id, created_dirs)
         clique = self.clique_id()
         # config consumed by the fabric daemon supervisor (imexd.cfg analog)
         with open(os.path.join(ddir, "fabricd.cfg.template"), "w") as f:
@@ -340,7 +462,13 @@ class ComputeDomainPlugin(dra.DRAPluginServicer):
 
     # -- channel device ------------------------------------------------------
 
-    def _prepare_channel(self, claim_msg, device_name: str, cfg: ComputeDomainChannelConfig):
+    def _prepare_channel(
+        self,
+        claim_msg,
+        device_name: str,
+        cfg: ComputeDomainChannelConfig,
+        created_dirs: List[str],
+    ):
         cd = self._get_cd_by_uid(cfg.domain_id)
         if cd is None:
             raise TransientError(f"ComputeDomain {cfg.domain_id} not found")
@@ -351,8 +479,7 @@ class ComputeDomainPlugin(dra.DRAPluginServicer):
             )
         self._label_node(cfg.domain_id)
         self._assert_domain_ready_on_node(cd)
-        ddir = self.domain_dir(cfg.domain_id)
-        os.makedirs(os.path.join(ddir, "shared"), exist_ok=True)
+        ddir = self._ensure_domain_dir(cfg.domain_id, created_dirs)
         if cfg.allocation_mode == ALLOCATION_MODE_ALL:
             channels = list(range(IMEX_CHANNEL_COUNT))
         else:
@@ -380,21 +507,7 @@ class ComputeDomainPlugin(dra.DRAPluginServicer):
             partition_index=0,
             device_nodes=[],
         )
-        self._assert_channel_not_allocated(claim_msg.uid, cfg.domain_id)
         return pd, cd_dev
-
-    def _assert_channel_not_allocated(self, claim_uid: str, domain_id: str) -> None:
-        """IMEX-channel exclusivity analog (ref device_state.go:729-757):
-        channel 0 of a domain may be prepared once per node per claim."""
-        cp = self.checkpoints.load()
-        for uid, pc in cp.claims().items():
-            if uid == claim_uid or pc is None:
-                continue
-            for d in pc.devices or []:
-                if d.type == "channel" and d.uuid == domain_id and d.partition_index == 0:
-                    raise PermanentError(
-                        f"channel 0 of domain {domain_id} already allocated to claim {uid}"
-                    )
 
     def _label_node(self, cd_uid: str) -> None:
         node = self.client.get_or_none("nodes", self.node_name)

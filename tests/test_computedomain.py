@@ -362,6 +362,256 @@ class TestCdPlugin:
         assert not os.path.exists(os.path.join(plugin.domains_dir, "dead-uid"))
 
 
+class TestCdPluginTwoPhase:
+    """Crash-recovery matrix for the 2-phase prepare protocol
+    (ref device_state.go:186-256,736-745; cleanup.go:117-125)."""
+
+    def _ready_clique(self, client, plugin, uid):
+        clique_id = plugin.clique_id()
+        client.create(
+            "computedomaincliques",
+            {"metadata": {"name": f"{uid}.{clique_id}"},
+             "daemons": [{"nodeName": "n1", "ipAddress": "10.0.0.1",
+                          "cliqueID": clique_id, "index": 0, "status": "Ready"}]},
+        )
+
+    def test_standard_spec_written_at_boot(self, cd_plugin):
+        client, lib, plugin = cd_plugin
+        spec = json.load(open(plugin.cdi.standard_spec_path()))
+        assert spec["devices"][0]["name"] == "all"
+        edits = spec["devices"][0]["containerEdits"]
+        node_paths = [n["path"] for n in edits["deviceNodes"]]
+        assert "/dev/kfd" in node_paths
+        assert any(p.startswith("/dev/dri/renderD") for p in node_paths)
+        assert "AMD_VISIBLE_DEVICES=void" in edits["env"]
+
+    def test_daemon_cdi_ids_include_standard_device(self, cd_plugin):
+        client, lib, plugin = cd_plugin
+        cd = make_cd(client)
+        uid = cd["metadata"]["uid"]
+        _mk_claim(client, "default", "dc", UID_CLAIM, "daemon-0",
+                  "ComputeDomainDaemonConfig", uid)
+        resp = plugin.node_prepare_resources(
+            dra.NodePrepareResourcesRequest(
+                claims=[dra.Claim(namespace="default", name="dc", uid=UID_CLAIM)]
+            ), None
+        )
+        ids = resp.claims[UID_CLAIM].devices[0].cdi_device_ids
+        assert ids[0] == plugin.cdi.qualified_name("all")
+        assert len(ids) == 2 and "claim-" in ids[1]
+
+    def test_channel_cdi_ids_have_no_standard_device(self, cd_plugin):
+        client, lib, plugin = cd_plugin
+        cd = make_cd(client)
+        uid = cd["metadata"]["uid"]
+        self._ready_clique(client, plugin, uid)
+        _mk_claim(client, "default", "wc", UID_CLAIM, "channel-0",
+                  "ComputeDomainChannelConfig", uid)
+        resp = plugin.node_prepare_resources(
+            dra.NodePrepareResourcesRequest(
+                claims=[dra.Claim(namespace="default", name="wc", uid=UID_CLAIM)]
+            ), None
+        )
+        ids = resp.claims[UID_CLAIM].devices[0].cdi_device_ids
+        assert len(ids) == 1 and "=all" not in ids[0]
+
+    def test_started_checkpointed_before_side_effects(self, cd_plugin):
+        """PrepareStarted is durable BEFORE the first side effect: a failure
+        inside device prepare leaves the annotated intent behind."""
+        client, lib, plugin = cd_plugin
+        cd = make_cd(client)
+        uid = cd["metadata"]["uid"]
+        _mk_claim(client, "default", "wc", UID_CLAIM, "channel-0",
+                  "ComputeDomainChannelConfig", uid)
+        # readiness gate never satisfied -> every attempt is transient
+        resp = plugin.node_prepare_resources(
+            dra.NodePrepareResourcesRequest(
+                claims=[dra.Claim(namespace="default", name="wc", uid=UID_CLAIM)]
+            ), None
+        )
+        assert "retry window exhausted" in resp.claims[UID_CLAIM].error
+        pc = plugin.checkpoints.load().get_claim(UID_CLAIM)
+        assert pc is not None and pc.state == "PrepareStarted"
+        assert pc.devices[0].type == "channel" and pc.devices[0].uuid == uid
+
+    def test_crash_between_phases_then_retry_completes(self, cd_plugin):
+        """Crash after side effects but before PrepareCompleted: the retry
+        re-enters, is not blocked by its own Started entry, and completes."""
+        client, lib, plugin = cd_plugin
+        cd = make_cd(client)
+        uid = cd["metadata"]["uid"]
+        self._ready_clique(client, plugin, uid)
+        _mk_claim(client, "default", "wc", UID_CLAIM, "channel-0",
+                  "ComputeDomainChannelConfig", uid)
+        real_write = plugin.cdi.write_claim_spec
+        calls = {"n": 0}
+
+        def crashy(uid_, devs):
+            calls["n"] += 1
+            out = real_write(uid_, devs)
+            if calls["n"] == 1:
+                raise RuntimeError("simulated crash after CDI write")
+            return out
+
+        plugin.cdi.write_claim_spec = crashy
+        plugin.retry_max_timeout = 0.05  # force the first request to fail
+        req = dra.NodePrepareResourcesRequest(
+            claims=[dra.Claim(namespace="default", name="wc", uid=UID_CLAIM)]
+        )
+        resp = plugin.node_prepare_resources(req, None)
+        assert resp.claims[UID_CLAIM].error != ""
+        assert plugin.checkpoints.load().get_claim(UID_CLAIM).state == "PrepareStarted"
+        # kubelet retries the whole request ("restart")
+        plugin.retry_max_timeout = 1.0
+        resp = plugin.node_prepare_resources(req, None)
+        assert resp.claims[UID_CLAIM].error == ""
+        assert plugin.checkpoints.load().get_claim(UID_CLAIM).state == "PrepareCompleted"
+
+    def test_started_claim_does_not_block_channel(self, cd_plugin):
+        """A claim stuck in PrepareStarted must NOT hold the channel: the
+        next claim wins and the stale one is reaped later
+        (ref device_state.go:736-745)."""
+        client, lib, plugin = cd_plugin
+        cd = make_cd(client)
+        uid = cd["metadata"]["uid"]
+        self._ready_clique(client, plugin, uid)
+        stale_uid = "cccccccc-0000-0000-0000-000000000003"
+        from k8s_dra_driver_gpu_amd.plugin.checkpoint import (
+            ClaimRef,
+            PreparedClaim,
+            PreparedDevice,
+        )
+
+        plugin.checkpoints.update(
+            lambda d: d.set_claim(
+                stale_uid,
+                PreparedClaim(
+                    state="PrepareStarted",
+                    claim=ClaimRef(namespace="default", name="dead", uid=stale_uid),
+                    devices=[PreparedDevice(type="channel", name="channel-0", uuid=uid)],
+                ),
+            )
+        )
+        _mk_claim(client, "default", "wc", UID_CLAIM, "channel-0",
+                  "ComputeDomainChannelConfig", uid)
+        resp = plugin.node_prepare_resources(
+            dra.NodePrepareResourcesRequest(
+                claims=[dra.Claim(namespace="default", name="wc", uid=UID_CLAIM)]
+            ), None
+        )
+        assert resp.claims[UID_CLAIM].error == ""
+
+    def test_losing_racer_rolled_back(self, cd_plugin):
+        """A concurrent (other-process) prepare completes between our phase 1
+        and commit: our commit must fail atomically AND our side effects
+        (CDI spec, checkpoint entry) must be rolled back."""
+        client, lib, plugin = cd_plugin
+        cd = make_cd(client)
+        uid = cd["metadata"]["uid"]
+        self._ready_clique(client, plugin, uid)
+        winner_uid = "dddddddd-0000-0000-0000-000000000004"
+        from k8s_dra_driver_gpu_amd.plugin.checkpoint import (
+            ClaimRef,
+            PreparedClaim,
+            PreparedDevice,
+        )
+
+        real_gate = plugin._assert_domain_ready_on_node
+        state = {"injected": False}
+
+        def gate_then_inject(cd_obj):
+            real_gate(cd_obj)
+            if not state["injected"]:
+                state["injected"] = True
+                # simulate another plugin process committing the channel
+                # between our phase 1 and our commit
+                plugin.checkpoints.update(
+                    lambda d: d.set_claim(
+                        winner_uid,
+                        PreparedClaim(
+                            state="PrepareCompleted",
+                            claim=ClaimRef(namespace="default", name="w", uid=winner_uid),
+                            devices=[
+                                PreparedDevice(type="channel", name="channel-0", uuid=uid)
+                            ],
+                        ),
+                    )
+                )
+
+        plugin._assert_domain_ready_on_node = gate_then_inject
+        _mk_claim(client, "default", "wc", UID_CLAIM, "channel-0",
+                  "ComputeDomainChannelConfig", uid)
+        resp = plugin.node_prepare_resources(
+            dra.NodePrepareResourcesRequest(
+                claims=[dra.Claim(namespace="default", name="wc", uid=UID_CLAIM)]
+            ), None
+        )
+        assert "already allocated" in resp.claims[UID_CLAIM].error
+        # rolled back: no checkpoint entry, no CDI spec for the loser
+        assert plugin.checkpoints.load().get_claim(UID_CLAIM) is None
+        assert not os.path.exists(plugin.cdi.claim_spec_path(UID_CLAIM))
+        # winner untouched
+        assert plugin.checkpoints.load().get_claim(winner_uid) is not None
+
+    def test_rollback_removes_unreferenced_created_dir(self, cd_plugin):
+        client, lib, plugin = cd_plugin
+        ddir = plugin.domain_dir("rollback-dom")
+        os.makedirs(os.path.join(ddir, "shared"))
+        from k8s_dra_driver_gpu_amd.plugin.checkpoint import ClaimRef, PreparedClaim
+
+        plugin.checkpoints.update(
+            lambda d: d.set_claim(
+                UID_CLAIM,
+                PreparedClaim(
+                    state="PrepareStarted",
+                    claim=ClaimRef(namespace="default", name="x", uid=UID_CLAIM),
+                ),
+            )
+        )
+        plugin.cdi.write_claim_spec(UID_CLAIM, [])
+        plugin._rollback_started(UID_CLAIM, [ddir])
+        assert not os.path.exists(ddir)
+        assert plugin.checkpoints.load().get_claim(UID_CLAIM) is None
+        assert not os.path.exists(plugin.cdi.claim_spec_path(UID_CLAIM))
+
+    def test_stale_started_claim_reaped_by_cleanup(self, cd_plugin):
+        """A PrepareStarted claim whose ResourceClaim is gone from the API
+        server is unprepared by the async cleanup pass (ref cleanup.go:117-125)."""
+        client, lib, plugin = cd_plugin
+        from k8s_dra_driver_gpu_amd.plugin.checkpoint import ClaimRef, PreparedClaim
+
+        plugin.checkpoints.update(
+            lambda d: d.set_claim(
+                UID_CLAIM,
+                PreparedClaim(
+                    state="PrepareStarted",
+                    claim=ClaimRef(namespace="default", name="ghost", uid=UID_CLAIM),
+                ),
+            )
+        )
+        assert plugin.cleanup_stale_claims() >= 1
+        assert plugin.checkpoints.load().get_claim(UID_CLAIM) is None
+
+    def test_unprepare_of_started_claim_is_clean(self, cd_plugin):
+        client, lib, plugin = cd_plugin
+        from k8s_dra_driver_gpu_amd.plugin.checkpoint import ClaimRef, PreparedClaim
+
+        plugin.checkpoints.update(
+            lambda d: d.set_claim(
+                UID_CLAIM,
+                PreparedClaim(
+                    state="PrepareStarted",
+                    claim=ClaimRef(namespace="default", name="x", uid=UID_CLAIM),
+                ),
+            )
+        )
+        resp = plugin.node_unprepare_resources(
+            dra.NodeUnprepareResourcesRequest(claims=[dra.Claim(uid=UID_CLAIM)]), None
+        )
+        assert resp.claims[UID_CLAIM].error == ""
+        assert plugin.checkpoints.load().get_claim(UID_CLAIM) is None
+
+
 # ---------------------------------------------------------------------------
 
 
