@@ -1,50 +1,66 @@
-"""GPU sharing managers: TimeSlicing and CPX spatial partitioning.
+"""GPU sharing managers: TimeSlicing and XCD spatial confinement.
 
 The reference implements sharing via ``nvidia-smi compute-policy
 --set-timeslice`` + compute mode, and an MPS control-daemon Deployment
-(``cmd/gpu-kubelet-plugin/sharing.go:75-178,214-436``).  MI355X has no MPS:
-the AMD-native sharing strategies are
+(``cmd/gpu-kubelet-plugin/sharing.go:75-178,214-436``).  MI355X has neither
+MPS nor a per-device timeslice knob, so the AMD-native strategies are:
 
-* **TimeSlicing** — amdgpu's hardware scheduler time-slices queues between
-  processes by default; the interval knob maps to the compute-scheduler
-  hysteresis setting (applied via amd-smi when available; recorded and
-  surfaced via env otherwise), and
-* **SpatialPartitioning** — confining a workload to a subset of XCDs, which
-  on ROCm is expressed per-process through ``HIP_VISIBLE_DEVICES``/CU masking
-  env rather than a control daemon.
+* **TimeSlicing** — amdgpu's hardware scheduler (HWS) time-slices compute
+  queues between processes by default; there is **no per-device interval
+  knob** on ROCm (verified against amd-smi 26.2.1: ``amd-smi set`` exposes
+  ``--compute-partition`` but no timeslice option; the KFD quantum is a
+  node-global module parameter).  The requested interval is therefore
+  surfaced to the container via the namespaced ``AMDDRA_SHARING`` env var
+  for observability, and sharing works because HWS already round-robins
+  queues — matching the reference's semantic of "multiple containers of one
+  claim share the GPU" (gpu-test2) without any device mutation.
 
-Both strategies therefore resolve to container env + (optionally) a
-device-level setting; the manager records what it applied so Unprepare can
-undo device-level settings.
+* **SpatialPartitioning** — confine the claim's containers to a subset of
+  the chip's XCDs using ``ROC_GLOBAL_CU_MASK``, the documented HIP/ROCclr
+  environment variable (present and parsed by this image's
+  ``libamdhip64.so``: "Setting CU mask 0x%s for hardware queue"): a hex
+  bitmask with one bit per CU applied to every queue the process creates.
+  MI355X has 256 CUs in 8 XCDs (32 CUs/XCD), so ``xcdCount=k`` emits a mask
+  with the low ``k*32`` bits set.  Hard partition isolation (separate
+  memory, separate device nodes) is the CPX partition path, not this.
+
+Every env var emitted here is either documented ROCm surface
+(``ROC_GLOBAL_CU_MASK``) or clearly ours (``AMDDRA_*``) — no pseudo-knobs.
 """
 
 from __future__ import annotations
 
 import logging
-import shutil
-import subprocess
 from typing import List, Optional
 
 from ..api.configs import DEFAULT_INTERVAL, GpuSharing
 
 logger = logging.getLogger("amddra.sharing")
 
-# amd-smi / kernel time-slice interval mapping (µs); Default leaves firmware
-# policy untouched (analog of ref sharing.go:188-230 interval model).
-TIMESLICE_US = {"Default": None, "Short": 500, "Medium": 2000, "Long": 5000}
+# MI355X: 8 XCDs x 32 CUs (aid: gfx950). Masks are sized for this geometry;
+# other CDNA parts would inject their own totals via the constructor.
+XCD_COUNT = 8
+CUS_PER_XCD = 32
+
+
+def cu_mask_hex(num_cus: int) -> str:
+    """Hex bitmask (``0x…``) with the low `num_cus` bits set — the format
+    ROC_GLOBAL_CU_MASK parses (one bit per CU, applied to all queues)."""
+    if num_cus <= 0:
+        num_cus = 1
+    return hex((1 << num_cus) - 1)
 
 
 class SharingManager:
     """Applies a sharing config for one prepared device and returns container
-    env edits; undoes device-level settings on remove."""
+    env edits. All strategies resolve to env only: ROCm has no per-device
+    sharing mutation to perform or undo (unlike the reference's
+    nvidia-smi/MPS paths), so prepare stays exec-free and remove() is a
+    no-op kept for interface parity with the reference's teardown hook."""
 
-    def __init__(self, amd_smi_path: str = ""):
-        self.amd_smi = amd_smi_path or shutil.which("amd-smi") or ""
-        # Device-level settings are idempotent: cache what is applied so
-        # repeated prepares don't re-exec amd-smi (the reference pays an
-        # nvidia-smi exec per prepare — this is one of our wins).
-        self._applied_timeslice: dict = {}
-        self._timeslice_unsupported = False
+    def __init__(self, xcds: int = XCD_COUNT, cus_per_xcd: int = CUS_PER_XCD):
+        self.xcds = xcds
+        self.cus_per_xcd = cus_per_xcd
 
     def apply(self, cfg, gpu=None, partition=None) -> List[str]:
         sharing: Optional[GpuSharing] = getattr(cfg, "sharing", None)
@@ -61,44 +77,26 @@ class SharingManager:
                 if sharing.time_slicing_config
                 else DEFAULT_INTERVAL
             )
+            # observability only: HWS time-slices by default; there is no
+            # per-device interval knob on ROCm (see module docstring)
             env.append(f"AMDDRA_SHARING=TimeSlicing:{interval}")
-            us = TIMESLICE_US.get(interval)
-            if us is not None and gpu is not None:
-                self._set_timeslice(gpu, us)
         elif sharing.is_spatial():
             sc = sharing.spatial_partitioning_config
             if sc and sc.xcd_count:
-                env.append(f"AMDDRA_SHARING=Spatial:xcd={sc.xcd_count}")
-                # CU mask: xcd_count/8 of the chip's CUs
-                env.append(f"HSA_CU_MASK_COUNT={sc.xcd_count * 32}")
+                n_xcds = min(sc.xcd_count, self.xcds)
             else:
                 pct = sc.default_xcd_percentage if sc else 100
-                env.append(f"AMDDRA_SHARING=Spatial:pct={pct}")
+                # XCD granularity: round down, at least one XCD
+                n_xcds = max(1, (pct * self.xcds) // 100)
+            env.append(f"AMDDRA_SHARING=Spatial:xcd={n_xcds}")
+            if n_xcds < self.xcds:
+                env.append(
+                    f"ROC_GLOBAL_CU_MASK={cu_mask_hex(n_xcds * self.cus_per_xcd)}"
+                )
+            # n_xcds == full chip: no mask needed (and none emitted)
         return env
 
-    def _set_timeslice(self, gpu, us: int) -> None:
-        """Best-effort device-level timeslice set via amd-smi (the
-        nvidia-smi-exec analog, ref nvlib.go:838-875). No-op when the tool or
-        the knob is unavailable (mock/CI)."""
-        if not self.amd_smi or self._timeslice_unsupported:
-            logger.debug("amd-smi unavailable; timeslice %dus recorded only", us)
-            return
-        if self._applied_timeslice.get(gpu.uuid) == us:
-            return
-        cmd = [self.amd_smi, "set", "--gpu", str(gpu.index), "--compute-partition-timeslice", str(us)]
-        try:
-            r = subprocess.run(cmd, capture_output=True, timeout=10, check=False)
-            if r.returncode != 0:
-                # knob not present on this platform/tool version: stop trying
-                self._timeslice_unsupported = True
-                logger.debug("amd-smi timeslice knob unsupported: %s", r.stderr[:200])
-            else:
-                self._applied_timeslice[gpu.uuid] = us
-        except Exception:
-            self._timeslice_unsupported = True
-            logger.debug("amd-smi timeslice set failed (non-fatal)", exc_info=True)
-
     def remove(self, prepared_device) -> None:
-        """Undo device-level sharing settings on unprepare (reset to Default
-        interval). Env edits die with the container; nothing else to do."""
+        """No device-level state to undo: sharing is env-only on ROCm (env
+        dies with the container)."""
         return None

@@ -102,6 +102,47 @@ class TestPrepareGpu:
         spec = json.load(open(cdi.claim_spec_path(UID1)))
         env_list = spec["devices"][0]["containerEdits"]["env"]
         assert "AMDDRA_SHARING=TimeSlicing:Long" in env_list
+        # honest sharing: no pseudo-knob env vars — every non-AMDDRA var we
+        # emit must be documented ROCm surface
+        for e in env_list:
+            assert e.split("=")[0].startswith(("AMDDRA_", "ROC_")), e
+
+    def _spatial_cfg(self, **spc):
+        return {
+            "apiVersion": APIVERSION,
+            "kind": "GpuConfig",
+            "sharing": {
+                "strategy": "SpatialPartitioning",
+                "spatialPartitioningConfig": spc,
+            },
+        }
+
+    def test_spatial_sharing_emits_documented_cu_mask(self, env):
+        """xcdCount=2 -> ROC_GLOBAL_CU_MASK (documented HIP env var) with the
+        low 64 bits set (2 XCDs x 32 CUs on MI355X)."""
+        _, _, cdi, _, ds = env
+        ds.prepare(claim(UID1, "gpu-0", configs=[self._spatial_cfg(xcdCount=2)]))
+        spec = json.load(open(cdi.claim_spec_path(UID1)))
+        env_list = spec["devices"][0]["containerEdits"]["env"]
+        assert "AMDDRA_SHARING=Spatial:xcd=2" in env_list
+        assert f"ROC_GLOBAL_CU_MASK={hex((1 << 64) - 1)}" in env_list
+
+    def test_spatial_full_chip_emits_no_mask(self, env):
+        _, _, cdi, _, ds = env
+        ds.prepare(claim(UID1, "gpu-0", configs=[self._spatial_cfg(xcdCount=8)]))
+        spec = json.load(open(cdi.claim_spec_path(UID1)))
+        env_list = spec["devices"][0]["containerEdits"]["env"]
+        assert not any(e.startswith("ROC_GLOBAL_CU_MASK") for e in env_list)
+
+    def test_spatial_percentage_rounds_to_xcd_granularity(self, env):
+        _, _, cdi, _, ds = env
+        ds.prepare(
+            claim(UID1, "gpu-0", configs=[self._spatial_cfg(defaultXcdPercentage=50)])
+        )
+        spec = json.load(open(cdi.claim_spec_path(UID1)))
+        env_list = spec["devices"][0]["containerEdits"]["env"]
+        assert "AMDDRA_SHARING=Spatial:xcd=4" in env_list
+        assert f"ROC_GLOBAL_CU_MASK={hex((1 << 128) - 1)}" in env_list
 
 
 class TestPreparePartition:
