@@ -8,10 +8,11 @@ cd "$(dirname "$0")/.."
 OUT=${1:-/tmp/amddra-asan}
 mkdir -p "$OUT"
 CXXFLAGS="-O1 -g -std=c++17 -fsanitize=address,undefined -fno-omit-frame-pointer -pthread"
-g++ $CXXFLAGS native/fabricd/fabricd.cpp -o "$OUT/fabricd" -ldl
+g++ $CXXFLAGS native/fabricd/fabricd.cpp -o "$OUT/fabricd" -ldl -lssl -lcrypto
 g++ $CXXFLAGS native/fabricd/fabricctl.cpp -o "$OUT/fabricctl"
 FABRICD_PATH="$OUT/fabricd" FABRICCTL_PATH="$OUT/fabricctl" \
   python -m pytest "tests/test_computedomain.py::TestFabricd" \
+                   "tests/test_computedomain.py::TestFabricdRobustness" \
                    "tests/test_computedomain.py::TestEightDaemonMesh" \
                    tests/test_e2e.py -q "$@"
 echo "sanitizer run clean"

@@ -42,6 +42,8 @@ int main(int argc, char** argv) {
         if (strcmp(argv[i], "-q") == 0) cmd = "STATUS";
         else if (strcmp(argv[i], "peers") == 0) cmd = "PEERS";
         else if (strcmp(argv[i], "probe") == 0) cmd = "PROBE";
+        else if (strcmp(argv[i], "metrics") == 0) cmd = "METRICS";
+        else if (strcmp(argv[i], "burn") == 0) cmd = "BURN";
         else if (strcmp(argv[i], "-p") == 0 && i + 1 < argc) port = atoi(argv[++i]);
         else if (strcmp(argv[i], "-H") == 0 && i + 1 < argc) host = argv[++i];
     }

@@ -9,14 +9,40 @@
 //
 // Protocol (newline-terminated text):
 //   peer port   (default 50000): "PING <domain>" -> "PONG <domain>"
-//   command port(default 50005): "STATUS" -> "READY" | "NOT_READY <n>/<m>"
+//                                plaintext, or mTLS when authMode=mtls
+//   command port(default 50005): "STATUS" -> "READY <n>/<m>" | "NOT_READY <n>/<m>"
 //                                "PEERS"  -> per-peer "host state"
 //                                "PROBE"  -> probe results (if run)
+//                                "METRICS"/"BURN" -> diagnostics
 // SIGUSR1 re-reads the nodes config (DNS-names mode: peers change without
 // daemon restart — ref compute-domain-daemon/main.go:384-431).
+// SIGTERM/SIGINT shut down cleanly (listeners are polled, server threads
+// joined) so the daemon is testable under ASan/LSan.
 //
-// Config: JSON file (-c), fields: domain, cliqueID, peerPort, commandPort,
-// nodesConfig (path to newline-separated peer hosts, relative to cfg dir).
+// Config: JSON file (-c). Fields (IMEX config-surface analogs,
+// ref templates/compute-domain-daemon-config.tmpl.cfg:84-218):
+//   domain, cliqueID, peerPort, commandPort, nodesConfig   — as before
+//   quorumPercent         readiness threshold: READY when
+//                         up*100 >= total*quorumPercent (default 100 =
+//                         all-peers; IMEX_WAIT_FOR_QUORUM analog, but as a
+//                         tunable degradation policy instead of
+//                         NONE/RECOVERY)
+//   disconnectedGraceSec  a peer that answered within this window still
+//                         counts as up (IMEX_NODE_DISCONNECTED_GRACE_TIME
+//                         analog; default 0 = immediate)
+//   reconnectBackoffMs /  per-peer exponential reconnect backoff base/cap
+//   reconnectBackoffMaxMs (default 250 ms -> 6 s, +-25% jitter — the
+//                         CD-daemon limiter shape, ref workqueue.go:61-63)
+//   authMode              "none" (default) or "mtls": peer-mesh
+//                         authentication+encryption via OpenSSL, both sides
+//                         verify against the CA (IMEX_ENABLE_AUTH_ENCRYPTION
+//                         + IMEX_AUTH_ENCRYPTION_MODE=SSL_TLS,
+//                         IMEX_AUTH_SOURCE=FILE analog). No hostname check
+//                         (IMEX_SECURITY_TARGET_OVERRIDE semantics: identity
+//                         is the CA-signed cert, peers move across IPs).
+//   tlsServerCert/tlsServerKey/tlsClientCert/tlsClientKey/tlsCa
+//                         PEM paths (relative to the config dir), used when
+//                         authMode=mtls (IMEX_SERVER_KEY/CERT/CERT_AUTH...)
 
 #include <arpa/inet.h>
 #include <dlfcn.h>
@@ -24,6 +50,9 @@
 #include <netdb.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <openssl/err.h>
+#include <openssl/ssl.h>
+#include <poll.h>
 #include <signal.h>
 #include <stdio.h>
 #include <stdlib.h>
@@ -35,12 +64,15 @@
 #include <chrono>
 #include <map>
 #include <mutex>
+#include <random>
 #include <string>
 #include <thread>
 #include <vector>
 
 static std::atomic<bool> g_reload{false};
 static std::atomic<bool> g_stop{false};
+static std::atomic<int> g_active_handlers{0};
+static std::atomic<long> g_reconnect_attempts{0};
 
 struct Config {
     std::string domain = "unknown";
@@ -50,6 +82,18 @@ struct Config {
     std::string nodes_config = "nodes.cfg";
     std::string cfg_dir = ".";
     bool gpu_probe = false;
+    int quorum_percent = 100;
+    int disconnected_grace_sec = 0;
+    int reconnect_backoff_ms = 250;
+    int reconnect_backoff_max_ms = 6000;
+    std::string auth_mode = "none";
+    std::string tls_server_cert, tls_server_key, tls_client_cert, tls_client_key, tls_ca;
+
+    bool mtls() const { return auth_mode == "mtls"; }
+    std::string rel(const std::string& p) const {
+        if (p.empty() || p[0] == '/') return p;
+        return cfg_dir + "/" + p;
+    }
 };
 
 // --- minimal JSON value extraction (flat string/int fields only) -----------
@@ -83,13 +127,26 @@ static bool load_config(const std::string& path, Config* cfg) {
     if (auto v = json_str(body, "peerPort"); !v.empty()) cfg->peer_port = atoi(v.c_str());
     if (auto v = json_str(body, "commandPort"); !v.empty()) cfg->command_port = atoi(v.c_str());
     if (auto v = json_str(body, "nodesConfig"); !v.empty()) cfg->nodes_config = v;
+    if (auto v = json_str(body, "quorumPercent"); !v.empty()) cfg->quorum_percent = atoi(v.c_str());
+    if (auto v = json_str(body, "disconnectedGraceSec"); !v.empty())
+        cfg->disconnected_grace_sec = atoi(v.c_str());
+    if (auto v = json_str(body, "reconnectBackoffMs"); !v.empty())
+        cfg->reconnect_backoff_ms = atoi(v.c_str());
+    if (auto v = json_str(body, "reconnectBackoffMaxMs"); !v.empty())
+        cfg->reconnect_backoff_max_ms = atoi(v.c_str());
+    if (auto v = json_str(body, "authMode"); !v.empty()) cfg->auth_mode = v;
+    cfg->tls_server_cert = json_str(body, "tlsServerCert");
+    cfg->tls_server_key = json_str(body, "tlsServerKey");
+    cfg->tls_client_cert = json_str(body, "tlsClientCert");
+    cfg->tls_client_key = json_str(body, "tlsClientKey");
+    cfg->tls_ca = json_str(body, "tlsCa");
+    if (cfg->quorum_percent < 1 || cfg->quorum_percent > 100) cfg->quorum_percent = 100;
     return true;
 }
 
 static std::vector<std::string> load_peers(const Config& cfg) {
     std::vector<std::string> peers;
-    std::string path = cfg.nodes_config[0] == '/' ? cfg.nodes_config
-                                                  : cfg.cfg_dir + "/" + cfg.nodes_config;
+    std::string path = cfg.rel(cfg.nodes_config);
     FILE* f = fopen(path.c_str(), "r");
     if (!f) return peers;
     char line[512];
@@ -107,12 +164,75 @@ static std::vector<std::string> load_peers(const Config& cfg) {
 struct PeerState {
     bool connected = false;
     std::chrono::steady_clock::time_point last_ok{};
+    int fail_streak = 0;
+    std::chrono::steady_clock::time_point next_attempt{};  // backoff gate
 };
 
 static std::mutex g_mu;
 static std::map<std::string, PeerState> g_peers;
 static std::string g_probe_report = "not run";
 static std::atomic<bool> g_probe_ok{true};
+
+// up = live connection, or last success within the disconnected-grace window
+static int count_up_locked(const Config& cfg) {
+    int up = 0;
+    auto now = std::chrono::steady_clock::now();
+    for (auto& [h, st] : g_peers) {
+        if (st.connected)
+            up++;
+        else if (cfg.disconnected_grace_sec > 0 &&
+                 st.last_ok.time_since_epoch().count() != 0 &&
+                 now - st.last_ok < std::chrono::seconds(cfg.disconnected_grace_sec))
+            up++;
+    }
+    return up;
+}
+
+static bool quorum_ready(const Config& cfg, int up, int total) {
+    return (long)up * 100 >= (long)total * cfg.quorum_percent;
+}
+
+// --- TLS -------------------------------------------------------------------
+static SSL_CTX* g_srv_ctx = nullptr;  // peer server (mTLS accept)
+static SSL_CTX* g_cli_ctx = nullptr;  // heartbeat client (mTLS connect)
+
+static SSL_CTX* make_ctx(const Config& cfg, bool server) {
+    SSL_CTX* ctx = SSL_CTX_new(server ? TLS_server_method() : TLS_client_method());
+    if (!ctx) return nullptr;
+    const std::string cert = cfg.rel(server ? cfg.tls_server_cert : cfg.tls_client_cert);
+    const std::string key = cfg.rel(server ? cfg.tls_server_key : cfg.tls_client_key);
+    const std::string ca = cfg.rel(cfg.tls_ca);
+    if (SSL_CTX_use_certificate_chain_file(ctx, cert.c_str()) != 1 ||
+        SSL_CTX_use_PrivateKey_file(ctx, key.c_str(), SSL_FILETYPE_PEM) != 1 ||
+        SSL_CTX_load_verify_locations(ctx, ca.c_str(), nullptr) != 1) {
+        fprintf(stderr, "fabricd: mTLS setup failed (%s / %s / %s): %s\n", cert.c_str(),
+                key.c_str(), ca.c_str(), ERR_error_string(ERR_get_error(), nullptr));
+        SSL_CTX_free(ctx);
+        return nullptr;
+    }
+    // both directions verify the peer certificate against the domain CA
+    SSL_CTX_set_verify(ctx, SSL_VERIFY_PEER | SSL_VERIFY_FAIL_IF_NO_PEER_CERT, nullptr);
+    SSL_CTX_set_min_proto_version(ctx, TLS1_2_VERSION);
+    return ctx;
+}
+
+// A peer-mesh connection: plain fd, or fd+SSL when mTLS is on.
+struct Conn {
+    int fd = -1;
+    SSL* ssl = nullptr;
+    bool valid() const { return fd >= 0; }
+    void close_all() {
+        if (ssl) {
+            SSL_shutdown(ssl);
+            SSL_free(ssl);
+            ssl = nullptr;
+        }
+        if (fd >= 0) {
+            close(fd);
+            fd = -1;
+        }
+    }
+};
 
 // --- sockets ---------------------------------------------------------------
 static int listen_on(int port) {
@@ -129,6 +249,14 @@ static int listen_on(int port) {
         return -1;
     }
     return fd;
+}
+
+// poll-gated accept: returns -1 on timeout/stop so the loop can exit cleanly
+static int accept_poll(int lfd) {
+    pollfd p{lfd, POLLIN, 0};
+    int r = poll(&p, 1, 200);
+    if (r <= 0 || !(p.revents & POLLIN)) return -1;
+    return accept(lfd, nullptr, nullptr);
 }
 
 static int connect_to(const std::string& host, int port, int timeout_ms) {
@@ -152,39 +280,43 @@ static int connect_to(const std::string& host, int port, int timeout_ms) {
     return fd;
 }
 
-static bool send_line(int fd, const std::string& line) {
-    // full-write loop: send() may return short counts or EINTR; a partial
-    // line would desync the peer protocol and flap connectivity
+static bool send_line(Conn& c, const std::string& line) {
+    // full-write loop: short counts / EINTR would desync the line protocol
     std::string msg = line + "\n";
     const char* p = msg.data();
     size_t n = msg.size();
     while (n > 0) {
-        ssize_t k = send(fd, p, n, MSG_NOSIGNAL);
-        if (k < 0) {
-            if (errno == EINTR) continue;
-            return false;
-        }
-        if (k == 0) return false;
+        ssize_t k;
+        if (c.ssl)
+            k = SSL_write(c.ssl, p, (int)n);
+        else
+            k = send(c.fd, p, n, MSG_NOSIGNAL);
+        if (k < 0 && !c.ssl && errno == EINTR) continue;
+        if (k <= 0) return false;
         p += k;
         n -= (size_t)k;
     }
     return true;
 }
 
-static std::string recv_line(int fd) {
+static std::string recv_line(Conn& c) {
     std::string out;
-    char c;
+    char ch;
     while (out.size() < 512) {
-        ssize_t n = recv(fd, &c, 1, 0);
-        if (n < 0 && errno == EINTR) continue;
+        ssize_t n;
+        if (c.ssl)
+            n = SSL_read(c.ssl, &ch, 1);
+        else
+            n = recv(c.fd, &ch, 1, 0);
+        if (n < 0 && !c.ssl && errno == EINTR) continue;
         if (n <= 0) break;
-        if (c == '\n') return out;
-        out.push_back(c);
+        if (ch == '\n') return out;
+        out.push_back(ch);
     }
     return out;
 }
 
-// --- peer service (answers PING) -------------------------------------------
+// --- peer service (answers PING; mTLS-wrapped when configured) --------------
 static void peer_server(const Config cfg) {
     int lfd = listen_on(cfg.peer_port);
     if (lfd < 0) {
@@ -193,21 +325,33 @@ static void peer_server(const Config cfg) {
         return;
     }
     while (!g_stop) {
-        sockaddr_in peer{};
-        socklen_t len = sizeof peer;
-        int fd = accept(lfd, (sockaddr*)&peer, &len);
+        int fd = accept_poll(lfd);
         if (fd < 0) continue;
-        std::thread([fd, cfg] {
+        g_active_handlers++;
+        std::thread([fd, &cfg] {
+            Conn c{fd, nullptr};
             timeval tv{5, 0};
             setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
+            setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof tv);
+            if (cfg.mtls()) {
+                c.ssl = SSL_new(g_srv_ctx);
+                SSL_set_fd(c.ssl, fd);
+                if (SSL_accept(c.ssl) != 1) {
+                    // unauthenticated peer: drop without answering
+                    c.close_all();
+                    g_active_handlers--;
+                    return;
+                }
+            }
             while (!g_stop) {
-                std::string line = recv_line(fd);
+                std::string line = recv_line(c);
                 if (line.rfind("PING ", 0) == 0)
-                    send_line(fd, "PONG " + line.substr(5));
+                    send_line(c, "PONG " + line.substr(5));
                 else
                     break;
             }
-            close(fd);
+            c.close_all();
+            g_active_handlers--;
         }).detach();
     }
     close(lfd);
@@ -215,7 +359,12 @@ static void peer_server(const Config cfg) {
 
 // --- heartbeat loop ---------------------------------------------------------
 static void heartbeat(const Config* cfg_ptr) {
-    std::map<std::string, int> conns;  // host -> fd
+    std::map<std::string, Conn> conns;  // host -> live connection
+    std::mt19937 rng{std::random_device{}()};
+    auto jittered = [&](int ms) {
+        std::uniform_real_distribution<double> d(0.75, 1.25);
+        return std::chrono::milliseconds((int)(ms * d(rng)));
+    };
     while (!g_stop) {
         std::vector<std::string> peers;
         {
@@ -223,35 +372,61 @@ static void heartbeat(const Config* cfg_ptr) {
             for (auto& [h, _] : g_peers) peers.push_back(h);
         }
         for (auto& host : peers) {
-            int fd = -1;
+            auto now = std::chrono::steady_clock::now();
+            Conn c;
             auto it = conns.find(host);
-            if (it != conns.end()) fd = it->second;
-            if (fd < 0) {
+            if (it != conns.end()) c = it->second;
+            if (!c.valid()) {
+                // reconnect gated by per-peer exponential backoff
+                {
+                    std::lock_guard<std::mutex> lk(g_mu);
+                    auto& st = g_peers[host];
+                    if (now < st.next_attempt) continue;
+                }
                 // peer entries may be "host" (cfg peer port) or "host:port"
                 std::string h = host;
                 int port = cfg_ptr->peer_port;
-                if (auto c = host.rfind(':'); c != std::string::npos) {
-                    h = host.substr(0, c);
-                    port = atoi(host.c_str() + c + 1);
+                if (auto col = host.rfind(':'); col != std::string::npos) {
+                    h = host.substr(0, col);
+                    port = atoi(host.c_str() + col + 1);
                 }
-                fd = connect_to(h, port, 1000);
-                if (fd >= 0) conns[host] = fd;
+                g_reconnect_attempts++;
+                int fd = connect_to(h, port, 1000);
+                if (fd >= 0) {
+                    c = Conn{fd, nullptr};
+                    if (cfg_ptr->mtls()) {
+                        c.ssl = SSL_new(g_cli_ctx);
+                        SSL_set_fd(c.ssl, fd);
+                        if (SSL_connect(c.ssl) != 1) c.close_all();
+                    }
+                }
+                if (c.valid()) conns[host] = c;
             }
             bool ok = false;
-            if (fd >= 0) {
-                if (send_line(fd, "PING " + cfg_ptr->domain)) {
-                    std::string resp = recv_line(fd);
+            if (c.valid()) {
+                if (send_line(c, "PING " + cfg_ptr->domain)) {
+                    std::string resp = recv_line(c);
                     ok = resp.rfind("PONG", 0) == 0;
                 }
                 if (!ok) {
-                    close(fd);
+                    c.close_all();
                     conns.erase(host);
                 }
             }
             std::lock_guard<std::mutex> lk(g_mu);
             auto& st = g_peers[host];
             st.connected = ok;
-            if (ok) st.last_ok = std::chrono::steady_clock::now();
+            if (ok) {
+                st.last_ok = std::chrono::steady_clock::now();
+                st.fail_streak = 0;
+                st.next_attempt = {};
+            } else {
+                long delay = (long)cfg_ptr->reconnect_backoff_ms << std::min(st.fail_streak, 8);
+                if (delay > cfg_ptr->reconnect_backoff_max_ms)
+                    delay = cfg_ptr->reconnect_backoff_max_ms;
+                st.fail_streak++;
+                st.next_attempt = std::chrono::steady_clock::now() + jittered((int)delay);
+            }
         }
         for (int i = 0; i < 10 && !g_stop && !g_reload; ++i)
             std::this_thread::sleep_for(std::chrono::milliseconds(100));
@@ -261,12 +436,18 @@ static void heartbeat(const Config* cfg_ptr) {
             std::lock_guard<std::mutex> lk(g_mu);
             std::map<std::string, PeerState> next;
             for (auto& h : fresh) next[h] = g_peers.count(h) ? g_peers[h] : PeerState{};
-            for (auto& [h, fd] : conns)
-                if (!next.count(h)) close(fd);
+            for (auto it2 = conns.begin(); it2 != conns.end();) {
+                if (!next.count(it2->first)) {
+                    it2->second.close_all();
+                    it2 = conns.erase(it2);
+                } else {
+                    ++it2;
+                }
+            }
             g_peers.swap(next);
         }
     }
-    for (auto& [h, fd] : conns) close(fd);
+    for (auto& [h, c] : conns) c.close_all();
 }
 
 // --- GPU probe (optional; dlopen the HIP probe library) ---------------------
@@ -315,29 +496,29 @@ static void command_server(const Config cfg) {
         return;
     }
     while (!g_stop) {
-        int fd = accept(lfd, nullptr, nullptr);
+        int fd = accept_poll(lfd);
         if (fd < 0) continue;
-        std::thread([fd] {
+        g_active_handlers++;
+        std::thread([fd, &cfg] {
+            Conn c{fd, nullptr};
             timeval tv{5, 0};
             setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
-            std::string line = recv_line(fd);
+            std::string line = recv_line(c);
             std::lock_guard<std::mutex> lk(g_mu);
             if (line == "STATUS") {
-                int up = 0, total = (int)g_peers.size();
-                for (auto& [h, st] : g_peers)
-                    if (st.connected) up++;
-                if (up == total && g_probe_ok)
-                    send_line(fd, "READY");
+                int up = count_up_locked(cfg), total = (int)g_peers.size();
+                bool ready = quorum_ready(cfg, up, total) && g_probe_ok;
+                std::string counts = std::to_string(up) + "/" + std::to_string(total);
+                if (ready)
+                    send_line(c, "READY " + counts);
                 else
-                    send_line(fd, "NOT_READY " + std::to_string(up) + "/" +
-                                      std::to_string(total) +
-                                      (g_probe_ok ? "" : " probe_failed"));
+                    send_line(c, "NOT_READY " + counts + (g_probe_ok ? "" : " probe_failed"));
             } else if (line == "PEERS") {
                 for (auto& [h, st] : g_peers)
-                    send_line(fd, h + " " + (st.connected ? "up" : "down"));
-                send_line(fd, "END");
+                    send_line(c, h + " " + (st.connected ? "up" : "down"));
+                send_line(c, "END");
             } else if (line == "PROBE") {
-                send_line(fd, g_probe_report);
+                send_line(c, g_probe_report);
             } else if (line == "BURN") {
                 // on-demand concurrent MFMA+HBM stress (the dcgmi diag -r
                 // analog); requires the probe library
@@ -345,32 +526,36 @@ static void command_server(const Config cfg) {
                                                              : "_libfabricprobe.so",
                                  RTLD_NOW);
                 if (!h) {
-                    send_line(fd, std::string("ERR probe library unavailable: ") + dlerror());
+                    send_line(c, std::string("ERR probe library unavailable: ") + dlerror());
                 } else {
                     auto burn = (int (*)(int, int, double*, double*))dlsym(h, "fp_burn");
                     double tf = 0, gb = 0;
                     if (burn && burn(0, 2000, &tf, &gb) == 0) {
                         char buf[128];
                         snprintf(buf, sizeof buf, "BURN_OK tflops=%.0f gbps=%.0f", tf, gb);
-                        send_line(fd, buf);
+                        send_line(c, buf);
                     } else {
-                        send_line(fd, "ERR burn failed");
+                        send_line(c, "ERR burn failed");
                     }
                 }
             } else if (line == "METRICS") {
-                int up = 0;
-                for (auto& [h, st] : g_peers)
-                    if (st.connected) up++;
-                send_line(fd, "# TYPE fabricd_peers gauge");
-                send_line(fd, "fabricd_peers " + std::to_string(g_peers.size()));
-                send_line(fd, "# TYPE fabricd_peers_connected gauge");
-                send_line(fd, "fabricd_peers_connected " + std::to_string(up));
-                send_line(fd, "# TYPE fabricd_probe_ok gauge");
-                send_line(fd, std::string("fabricd_probe_ok ") + (g_probe_ok ? "1" : "0"));
+                int up = count_up_locked(cfg);
+                send_line(c, "# TYPE fabricd_peers gauge");
+                send_line(c, "fabricd_peers " + std::to_string(g_peers.size()));
+                send_line(c, "# TYPE fabricd_peers_connected gauge");
+                send_line(c, "fabricd_peers_connected " + std::to_string(up));
+                send_line(c, "# TYPE fabricd_probe_ok gauge");
+                send_line(c, std::string("fabricd_probe_ok ") + (g_probe_ok ? "1" : "0"));
+                send_line(c, "# TYPE fabricd_reconnect_attempts_total counter");
+                send_line(c, "fabricd_reconnect_attempts_total " +
+                                 std::to_string(g_reconnect_attempts.load()));
+                send_line(c, "# TYPE fabricd_quorum_percent gauge");
+                send_line(c, "fabricd_quorum_percent " + std::to_string(cfg.quorum_percent));
             } else {
-                send_line(fd, "ERR unknown command");
+                send_line(c, "ERR unknown command");
             }
-            close(fd);
+            c.close_all();
+            g_active_handlers--;
         }).detach();
     }
     close(lfd);
@@ -389,6 +574,12 @@ int main(int argc, char** argv) {
     if (const char* p = getenv("FABRICD_GPU_PROBE"); p && strcmp(p, "1") == 0)
         cfg.gpu_probe = true;
 
+    if (cfg.mtls()) {
+        g_srv_ctx = make_ctx(cfg, true);
+        g_cli_ctx = make_ctx(cfg, false);
+        if (!g_srv_ctx || !g_cli_ctx) return 1;
+    }
+
     signal(SIGUSR1, [](int) { g_reload = true; });
     signal(SIGTERM, [](int) { g_stop = true; });
     signal(SIGINT, [](int) { g_stop = true; });
@@ -399,9 +590,11 @@ int main(int argc, char** argv) {
         std::lock_guard<std::mutex> lk(g_mu);
         for (auto& h : peers) g_peers[h] = PeerState{};
     }
-    fprintf(stderr, "fabricd: domain=%s clique=%s peers=%zu ports=%d/%d\n",
+    fprintf(stderr,
+            "fabricd: domain=%s clique=%s peers=%zu ports=%d/%d quorum=%d%% grace=%ds auth=%s\n",
             cfg.domain.c_str(), cfg.clique_id.c_str(), g_peers.size(), cfg.peer_port,
-            cfg.command_port);
+            cfg.command_port, cfg.quorum_percent, cfg.disconnected_grace_sec,
+            cfg.auth_mode.c_str());
 
     if (cfg.gpu_probe) run_gpu_probe();
 
@@ -409,6 +602,15 @@ int main(int argc, char** argv) {
     std::thread t2(command_server, cfg);
     std::thread t3(heartbeat, &cfg);
     while (!g_stop) std::this_thread::sleep_for(std::chrono::milliseconds(200));
-    // threads hold blocking accepts; exit hard (the supervisor owns lifecycle)
-    _exit(0);
+    // clean shutdown: accept loops are polled (exit within 200 ms), the
+    // heartbeat observes g_stop; give detached handlers (5 s recv timeout)
+    // a bounded drain so ASan/LSan see a quiet process
+    t1.join();
+    t2.join();
+    t3.join();
+    for (int i = 0; i < 60 && g_active_handlers > 0; ++i)
+        std::this_thread::sleep_for(std::chrono::milliseconds(100));
+    if (g_srv_ctx) SSL_CTX_free(g_srv_ctx);
+    if (g_cli_ctx) SSL_CTX_free(g_cli_ctx);
+    return 0;
 }

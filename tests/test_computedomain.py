@@ -711,7 +711,7 @@ needs_fabricd = pytest.mark.skipif(
 
 @needs_fabricd
 class TestFabricd:
-    def _write_cfg(self, d, peer_port, cmd_port, peers):
+    def _write_cfg(self, d, peer_port, cmd_port, peers, **extra):
         cfg = {
             "domain": "test-dom",
             "cliqueID": "h.0",
@@ -719,6 +719,7 @@ class TestFabricd:
             "commandPort": cmd_port,
             "nodesConfig": "nodes.cfg",
         }
+        cfg.update(extra)
         with open(os.path.join(d, "fabricd.cfg"), "w") as f:
             json.dump(cfg, f)
         with open(os.path.join(d, "nodes.cfg"), "w") as f:
@@ -750,7 +751,7 @@ class TestFabricd:
         ]
         try:
             ok = wait_for(
-                lambda: self._status(c1) == "READY" and self._status(c2) == "READY",
+                lambda: self._status(c1).startswith("READY") and self._status(c2).startswith("READY"),
                 timeout=30.0,
                 interval=0.3,
             )
@@ -772,7 +773,7 @@ class TestFabricd:
         ]
         try:
             ok = wait_for(
-                lambda: self._status(c1) == "READY" and self._status(c2) == "READY",
+                lambda: self._status(c1).startswith("READY") and self._status(c2).startswith("READY"),
                 timeout=30.0,
                 interval=0.3,
             )
@@ -799,7 +800,7 @@ class TestFabricd:
         self._write_cfg(d1, p1, c1, [])
         proc = subprocess.Popen([FABRICD, "-c", os.path.join(d1, "fabricd.cfg")])
         try:
-            ok = wait_for(lambda: self._status(c1) == "READY", timeout=10.0, interval=0.2)
+            ok = wait_for(lambda: self._status(c1).startswith("READY"), timeout=10.0, interval=0.2)
             assert ok  # no peers -> READY
             with open(os.path.join(d1, "nodes.cfg"), "w") as f:
                 f.write("127.0.0.1:1\n")  # unreachable peer
@@ -811,6 +812,185 @@ class TestFabricd:
         finally:
             proc.kill()
             proc.wait(timeout=5)
+
+
+@needs_fabricd
+class TestFabricdRobustness:
+    """Quorum readiness, flap tolerance, clean shutdown, reconnect backoff
+    and mTLS (VERDICT round-1 item 7; IMEX config-surface parity per
+    templates/compute-domain-daemon-config.tmpl.cfg:84-218)."""
+
+    _write_cfg = TestFabricd._write_cfg
+    _status = TestFabricd._status
+
+    def _metrics(self, port):
+        from k8s_dra_driver_gpu_amd.daemon.process import default_fabricctl_path
+
+        out = subprocess.run(
+            [default_fabricctl_path(), "metrics", "-p", str(port)],
+            capture_output=True, text=True, timeout=10,
+        )
+        return out.stdout
+
+    def test_quorum_ready_with_one_dead_peer(self, tmp_path):
+        """quorumPercent=50: 1 of 2 peers dead -> domain degrades to READY,
+        not forever-NotReady (the IMEX quorum RECOVERY analog)."""
+        da, db = str(tmp_path / "a"), str(tmp_path / "b")
+        os.makedirs(da), os.makedirs(db)
+        pa, ca, pb, cb = _free_port(), _free_port(), _free_port(), _free_port()
+        dead = _free_port()  # nothing listens here
+        self._write_cfg(da, pa, ca, [f"127.0.0.1:{pb}", f"127.0.0.1:{dead}"],
+                        quorumPercent=50)
+        self._write_cfg(db, pb, cb, [f"127.0.0.1:{pa}"])
+        procs = [subprocess.Popen([FABRICD, "-c", os.path.join(d, "fabricd.cfg")])
+                 for d in (da, db)]
+        try:
+            ok = wait_for(lambda: self._status(ca).startswith("READY 1/2"),
+                          timeout=20.0, interval=0.3)
+            assert ok, self._status(ca)
+        finally:
+            for p in procs:
+                p.kill()
+                p.wait(timeout=5)
+
+    def test_all_peers_policy_not_ready_with_dead_peer(self, tmp_path):
+        """Default (quorumPercent=100) keeps the strict all-peers semantic."""
+        da = str(tmp_path / "a")
+        os.makedirs(da)
+        pa, ca = _free_port(), _free_port()
+        self._write_cfg(da, pa, ca, [f"127.0.0.1:{_free_port()}"])
+        proc = subprocess.Popen([FABRICD, "-c", os.path.join(da, "fabricd.cfg")])
+        try:
+            wait_for(lambda: self._status(ca).startswith("NOT_READY"), timeout=10.0)
+            assert self._status(ca).startswith("NOT_READY 0/1")
+        finally:
+            proc.kill()
+            proc.wait(timeout=5)
+
+    def test_clean_shutdown_on_sigterm(self, tmp_path):
+        """SIGTERM exits 0 via joined threads (no _exit) — the property the
+        ASan tier needs for leak checking."""
+        import signal as sig
+
+        da = str(tmp_path / "a")
+        os.makedirs(da)
+        pa, ca = _free_port(), _free_port()
+        self._write_cfg(da, pa, ca, [])
+        proc = subprocess.Popen([FABRICD, "-c", os.path.join(da, "fabricd.cfg")])
+        try:
+            wait_for(lambda: self._status(ca).startswith("READY"), timeout=10.0)
+            proc.send_signal(sig.SIGTERM)
+            assert proc.wait(timeout=15) == 0
+        finally:
+            if proc.poll() is None:
+                proc.kill()
+                proc.wait(timeout=5)
+
+    def test_reconnect_backoff_bounds_attempts(self, tmp_path):
+        """A dead peer is retried with exponential backoff, not hammered
+        every heartbeat tick."""
+        da = str(tmp_path / "a")
+        os.makedirs(da)
+        pa, ca = _free_port(), _free_port()
+        self._write_cfg(da, pa, ca, [f"127.0.0.1:{_free_port()}"],
+                        reconnectBackoffMs=2000, reconnectBackoffMaxMs=8000)
+        proc = subprocess.Popen([FABRICD, "-c", os.path.join(da, "fabricd.cfg")])
+        try:
+            wait_for(lambda: self._status(ca).startswith("NOT_READY"), timeout=10.0)
+            time.sleep(5.0)
+            m = self._metrics(ca)
+            attempts = int(
+                [l for l in m.splitlines() if l.startswith("fabricd_reconnect_attempts_total")][0]
+                .split()[-1]
+            )
+            # without backoff the 1 s heartbeat would make ~6+ attempts;
+            # with a 2 s base doubling to 8 s we see at most ~4
+            assert 1 <= attempts <= 4, m
+            assert "fabricd_quorum_percent 100" in m
+        finally:
+            proc.kill()
+            proc.wait(timeout=5)
+
+    def test_disconnected_grace_keeps_peer_up(self, tmp_path):
+        """disconnectedGraceSec: a freshly-dead peer still counts as up
+        inside the grace window (IMEX_NODE_DISCONNECTED_GRACE_TIME analog)."""
+        da, db = str(tmp_path / "a"), str(tmp_path / "b")
+        os.makedirs(da), os.makedirs(db)
+        pa, ca, pb, cb = _free_port(), _free_port(), _free_port(), _free_port()
+        self._write_cfg(da, pa, ca, [f"127.0.0.1:{pb}"], disconnectedGraceSec=30)
+        self._write_cfg(db, pb, cb, [f"127.0.0.1:{pa}"])
+        procs = [subprocess.Popen([FABRICD, "-c", os.path.join(d, "fabricd.cfg")])
+                 for d in (da, db)]
+        try:
+            ok = wait_for(lambda: self._status(ca).startswith("READY 1/1"), timeout=20.0)
+            assert ok, self._status(ca)
+            procs[1].kill()
+            procs[1].wait(timeout=5)
+            time.sleep(2.5)  # heartbeat notices the drop
+            # still READY: inside the 30 s grace window
+            assert self._status(ca).startswith("READY"), self._status(ca)
+        finally:
+            for p in procs:
+                if p.poll() is None:
+                    p.kill()
+                    p.wait(timeout=5)
+
+    # -- mTLS ---------------------------------------------------------------
+
+    def _gen_certs(self, d):
+        """CA + one node cert signed by it (IMEX_AUTH_SOURCE=FILE analog)."""
+        def run(*args):
+            subprocess.run(list(args), check=True, capture_output=True, cwd=d)
+
+        run("openssl", "req", "-x509", "-newkey", "rsa:2048", "-keyout", "ca.key",
+            "-out", "ca.crt", "-days", "2", "-nodes", "-subj", "/CN=fabric-ca")
+        run("openssl", "req", "-newkey", "rsa:2048", "-keyout", "node.key",
+            "-out", "node.csr", "-nodes", "-subj", "/CN=fabric-node")
+        run("openssl", "x509", "-req", "-in", "node.csr", "-CA", "ca.crt",
+            "-CAkey", "ca.key", "-CAcreateserial", "-out", "node.crt", "-days", "2")
+
+    def _mtls_cfg(self, certs_dir):
+        return dict(
+            authMode="mtls",
+            tlsServerCert=os.path.join(certs_dir, "node.crt"),
+            tlsServerKey=os.path.join(certs_dir, "node.key"),
+            tlsClientCert=os.path.join(certs_dir, "node.crt"),
+            tlsClientKey=os.path.join(certs_dir, "node.key"),
+            tlsCa=os.path.join(certs_dir, "ca.crt"),
+        )
+
+    def test_mtls_mesh_reaches_ready(self, tmp_path):
+        certs = str(tmp_path / "certs")
+        os.makedirs(certs)
+        self._gen_certs(certs)
+        da, db = str(tmp_path / "a"), str(tmp_path / "b")
+        os.makedirs(da), os.makedirs(db)
+        pa, ca, pb, cb = _free_port(), _free_port(), _free_port(), _free_port()
+        self._write_cfg(da, pa, ca, [f"127.0.0.1:{pb}"], **self._mtls_cfg(certs))
+        self._write_cfg(db, pb, cb, [f"127.0.0.1:{pa}"], **self._mtls_cfg(certs))
+        procs = [subprocess.Popen([FABRICD, "-c", os.path.join(d, "fabricd.cfg")])
+                 for d in (da, db)]
+        try:
+            ok = wait_for(
+                lambda: self._status(ca).startswith("READY") and self._status(cb).startswith("READY"),
+                timeout=25.0, interval=0.3,
+            )
+            assert ok, f"{self._status(ca)} / {self._status(cb)}"
+            # an unauthenticated plaintext client gets no PONG from the
+            # mTLS peer port
+            s = socket.create_connection(("127.0.0.1", pa), timeout=3)
+            s.sendall(b"PING test-dom\n")
+            s.settimeout(3)
+            try:
+                data = s.recv(64)
+            except (socket.timeout, ConnectionError):
+                data = b""
+            s.close()
+            assert not data.startswith(b"PONG")
+        finally:
+            for p in procs:
+                p.kill()
+                p.wait(timeout=5)
 
 
 class TestProcessManager:
