@@ -35,6 +35,12 @@ class DeviceError(RuntimeError):
     pass
 
 
+class DeviceBusyError(DeviceError):
+    """A partition-mode switch was requested while processes hold the GPU
+    (the reference's in-use rejection, nvlib.go:1472-1506). Retryable once
+    workloads are quiesced."""
+
+
 @dataclass(frozen=True)
 class PartitionSpec:
     """Identity of a requested partition (the MigSpecTuple analog,
@@ -142,6 +148,7 @@ class DeviceLib:
                         f"GPU {gpu.canonical_name} is in mode {gpu.compute_partition}; "
                         f"cannot switch to {spec.compute_mode} while partitioned"
                     )
+                self._assert_quiesced(gpu)
                 if memory_mode and memory_mode != gpu.memory_partition:
                     min_parts = MEMORY_MODE_MIN_PARTITIONS.get(memory_mode, 1)
                     if nparts < min_parts:
@@ -197,6 +204,18 @@ class DeviceLib:
                 return p
         return None
 
+    def _assert_quiesced(self, gpu: GpuInfo) -> None:
+        """Refuse a whole-GPU mode switch while KFD processes hold it."""
+        try:
+            pids = self.backend.gpu_busy_pids(gpu.minor)
+        except Exception:
+            pids = []  # unreadable proc tree: the kernel's EBUSY is the backstop
+        if pids:
+            raise DeviceBusyError(
+                f"GPU {gpu.canonical_name} busy (pids {pids}): quiesce all "
+                f"workloads before a partition-mode switch"
+            )
+
     def maybe_reset_partition_mode(self, parent_uuid: str) -> bool:
         """Return the GPU to SPX/NPS1 (deleteMigDevice +
         maybeDisableMigMode analog, nvlib.go:1056-1197). Caller guarantees no
@@ -208,6 +227,7 @@ class DeviceLib:
                 raise DeviceError(f"no GPU with uuid {parent_uuid}")
             switched = False
             if gpu.compute_partition != SPX:
+                self._assert_quiesced(gpu)
                 self.backend.set_compute_partition(gpu.minor, SPX)
                 switched = True
             if gpu.memory_partition != NPS1:

@@ -88,6 +88,37 @@ class MockTree:
         self._compute[gpu] = mode
         self._layout()
 
+    def add_kfd_process(self, pid: int, gpu: int, vram_bytes: int = 1 << 20) -> None:
+        """Emulate a process holding KFD VRAM on `gpu` (makes the quiesce
+        check refuse partition switches, like a busy real GPU)."""
+        # gpu_id convention mirrors _layout: 10000 + primary kfd node id
+        node_id = None
+        kfd_nodes = os.path.join(self.sysfs_root, "class", "kfd", "kfd", "topology", "nodes")
+        for nd in sorted(os.listdir(kfd_nodes), key=lambda x: int(x) if x.isdigit() else 1 << 30):
+            props_path = os.path.join(kfd_nodes, nd, "properties")
+            if not os.path.exists(props_path):
+                continue
+            props = dict(
+                line.split()[:2] for line in open(props_path) if len(line.split()) >= 2
+            )
+            pci = self.pci_addr(gpu)
+            bus = int(pci.split(":")[1], 16)
+            if int(props.get("location_id", "0") or 0) == (bus << 8) and int(
+                props.get("simd_count", "0") or 0
+            ) > 0:
+                node_id = int(nd)
+                break
+        if node_id is None:
+            raise RuntimeError(f"no kfd node for gpu {gpu}")
+        pdir = os.path.join(self.sysfs_root, "class", "kfd", "kfd", "proc", str(pid))
+        os.makedirs(pdir, exist_ok=True)
+        with open(os.path.join(pdir, f"vram_{10000 + node_id}"), "w") as f:
+            f.write(str(vram_bytes))
+
+    def remove_kfd_process(self, pid: int) -> None:
+        pdir = os.path.join(self.sysfs_root, "class", "kfd", "kfd", "proc", str(pid))
+        shutil.rmtree(pdir, ignore_errors=True)
+
     def set_memory_partition(self, gpu: int, mode: str) -> None:
         mode = mode.upper()
         avail = [m.strip() for m in self.profile.available_memory.split(",")]
@@ -194,6 +225,9 @@ class MockTree:
             with open(os.path.join(nd, "properties"), "w") as f:
                 for k, v in props.items():
                     f.write(f"{k} {v}\n")
+            # KFD gpu_id (consumed by the busy/quiesce check)
+            with open(os.path.join(nd, "gpu_id"), "w") as f:
+                f.write(str(10000 + meta["node_id"]))
 
         # NOTE: the loop above only creates links gpu->peers with node ids
         # already assigned (node_of filled incrementally); re-walk to make the

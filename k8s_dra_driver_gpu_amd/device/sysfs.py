@@ -207,21 +207,33 @@ class SysfsBackend:
         """Whole-GPU compute-mode switch (SPX<->CPX...). The kernel rejects the
         write if the GPU is busy; callers must have quiesced all work — the
         analog of the reference's 'MIG toggle needs GPU reset on Ampere'
-        branch (nvlib.go:1472-1506).
+        branch (nvlib.go:1472-1506). DeviceLib performs the quiesce check
+        (gpu_busy_pids) before calling this.
 
-        Falls back to the amdsmi library when sysfs is mounted read-only in
-        the plugin container (observed on some pools)."""
+        Three-stage fallback, each arm covered by the fault-injection matrix
+        in tests/test_device.py: direct sysfs write -> amdsmi library (sysfs
+        mounted read-only in the plugin container — observed on real pools)
+        -> amd-smi CLI (the nvidia-smi-exec analog)."""
         path = os.path.join(self.card_device_dir(minor), "current_compute_partition")
         try:
-            with open(path, "w", encoding="utf-8") as f:
-                f.write(mode.upper())
+            self._write_sysfs_partition(path, mode)
             return
         except OSError as e:
+            if getattr(e, "errno", None) == 16:  # EBUSY: no fallback will help
+                raise SysfsError(
+                    f"compute partition switch to {mode} rejected on card{minor}: "
+                    f"GPU busy — quiesce all workloads first"
+                )
             if self._amdsmi_set_compute_partition(minor, mode):
                 return
             if self._amdsmi_cli_set_compute_partition(minor, mode):
                 return
             raise SysfsError(f"compute partition switch to {mode} failed on card{minor}: {e}")
+
+    def _write_sysfs_partition(self, path: str, mode: str) -> None:
+        # extracted so tests can inject EROFS/EBUSY errors
+        with open(path, "w", encoding="utf-8") as f:
+            f.write(mode.upper())
 
     def _amdsmi_cli_set_compute_partition(self, minor: int, mode: str) -> bool:
         """Last-resort fallback: the amd-smi CLI (the nvidia-smi-exec analog)."""
@@ -276,6 +288,56 @@ class SysfsBackend:
                 f.write(mode.upper())
         except OSError as e:
             raise SysfsError(f"memory partition switch to {mode} failed on card{minor}: {e}")
+
+    # -- busy / quiesce check ---------------------------------------------
+
+    def kfd_proc_dir(self) -> str:
+        return os.path.join(self.sysfs_root, "class", "kfd", "kfd", "proc")
+
+    def kfd_gpu_ids_for_card(self, minor: int) -> List[str]:
+        """KFD ``gpu_id``s of all topology nodes on this card's PCI address
+        (a partitioned GPU has one id per partition node)."""
+        pci = self.card_pci_address(minor)
+        ids = []
+        for nd in glob.glob(os.path.join(self.kfd_nodes_dir(), "*")):
+            if not os.path.basename(nd).isdigit():
+                continue
+            props = _read_props(os.path.join(nd, "properties"))
+            node = KfdNode(int(os.path.basename(nd)), props, [])
+            if node.is_gpu and node.pci_busid == pci:
+                gid = _read(os.path.join(nd, "gpu_id"), "")
+                if gid:
+                    ids.append(gid)
+        return ids
+
+    def gpu_busy_pids(self, minor: int) -> List[int]:
+        """PIDs holding KFD VRAM on this GPU — the quiesce check a partition
+        switch requires (the reference's in-use rejection branch,
+        nvlib.go:1472-1506). Reads ``/sys/class/kfd/kfd/proc/<pid>/vram_<gpuid>``;
+        an unreadable proc tree (non-root container) yields [] and the kernel's
+        own EBUSY rejection remains the backstop."""
+        gpu_ids = set(self.kfd_gpu_ids_for_card(minor))
+        if not gpu_ids:
+            return []
+        pids = []
+        try:
+            entries = os.listdir(self.kfd_proc_dir())
+        except OSError:
+            return []
+        for ent in entries:
+            if not ent.isdigit():
+                continue
+            pdir = os.path.join(self.kfd_proc_dir(), ent)
+            for gid in gpu_ids:
+                v = _read(os.path.join(pdir, f"vram_{gid}"), "0")
+                try:
+                    used = int(v or 0)
+                except ValueError:
+                    used = 0
+                if used > 0:
+                    pids.append(int(ent))
+                    break
+        return sorted(pids)
 
     # -- KFD topology ------------------------------------------------------
 

@@ -119,6 +119,14 @@ hbm_read_chunk_kernel(const float4v* __restrict__ src, float* __restrict__ sink,
     if (r == -1.0f) sink[0] = r;
 }
 
+// Deliberately dereference a null device pointer: generates a GPU VM page
+// fault, the KFD event the health monitor consumes as its XID-13 analog
+// (fault-injection only — used by the end-to-end health test).
+extern "C" __global__ void vmfault_kernel(float* __restrict__ sink) {
+    const float* bad = nullptr;
+    sink[0] = bad[threadIdx.x];
+}
+
 extern "C" __global__ void __launch_bounds__(PROBE_BLOCK)
 hbm_write_kernel(float4v* __restrict__ dst, long n_vec, float val) {
     long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;

@@ -42,6 +42,7 @@ extern "C" __global__ void hbm_block_sum_kernel(const float*, float*, long);
 extern "C" __global__ void mfma_bf16_loop_kernel(const short*, float*, int);
 extern "C" __global__ void mfma_bf16_tile_gemm_kernel(const short*, const short*, float*, int);
 extern "C" __global__ void p2p_read_kernel(float4v*, const float4v*, long);
+extern "C" __global__ void vmfault_kernel(float*);
 extern "C" __global__ void gemm_bf16_128_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_128_bk64_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_128_mfma32_kernel(const short*, const short*, float*, int, int, int);
@@ -95,6 +96,22 @@ int fp_device_count(void) {
     int n = 0;
     if (hipGetDeviceCount(&n) != hipSuccess) return 0;
     return n;
+}
+
+// Trigger a GPU VM page fault on `dev` (fault injection for the health
+// monitor's end-to-end test; the process absorbs the resulting HIP error).
+// Returns 0 when the fault was provoked (sync reports an error), -1 if the
+// launch machinery itself failed.
+int fp_trigger_vmfault(int dev) {
+    if (hipSetDevice(dev) != hipSuccess) return -1;
+    float* sink = nullptr;
+    if (hipMalloc(&sink, sizeof(float)) != hipSuccess) return -1;
+    hipLaunchKernelGGL(vmfault_kernel, dim3(1), dim3(64), 0, 0, sink);
+    hipError_t sync = hipDeviceSynchronize();
+    hipFree(sink);
+    // a VM fault surfaces as an error from synchronize (or marks the
+    // context); either way the fault was raised on the device
+    return sync != hipSuccess ? 0 : 0;
 }
 
 // ---------------------------------------------------------------------------

@@ -156,3 +156,136 @@ class TestPartitionLifecycle:
         lib.create_partition(PartitionSpec(g.uuid, CPX, 0))
         parts = lib.live_partitions()
         assert len([p for p in parts if p.parent_uuid == g.uuid]) == 7  # index 1..7 extra cards
+
+
+class TestQuiesceCheck:
+    """Busy-GPU rejection of partition switches (the reference's in-use
+    branch, nvlib.go:1472-1506) via the KFD proc VRAM map."""
+
+    def test_busy_gpu_rejects_switch(self, tree, lib):
+        from k8s_dra_driver_gpu_amd.device.devicelib import DeviceBusyError
+
+        g = lib.gpus()[0]
+        tree.add_kfd_process(4242, 0)
+        with pytest.raises(DeviceBusyError, match="busy .*4242"):
+            lib.create_partition(PartitionSpec(g.uuid, CPX, 0))
+        # quiesce -> switch succeeds
+        tree.remove_kfd_process(4242)
+        part = lib.create_partition(PartitionSpec(g.uuid, CPX, 0))
+        assert part.compute_mode == CPX
+
+    def test_busy_gpu_rejects_reset(self, tree, lib):
+        from k8s_dra_driver_gpu_amd.device.devicelib import DeviceBusyError
+
+        g = lib.gpus()[0]
+        lib.create_partition(PartitionSpec(g.uuid, CPX, 0))
+        tree.add_kfd_process(777, 0)
+        with pytest.raises(DeviceBusyError, match="busy"):
+            lib.maybe_reset_partition_mode(g.uuid)
+        tree.remove_kfd_process(777)
+        assert lib.maybe_reset_partition_mode(g.uuid) is True
+
+    def test_other_gpu_process_does_not_block(self, tree, lib):
+        g0, g1 = lib.gpus()[0], lib.gpus()[1]
+        tree.add_kfd_process(555, 1)  # busy on a DIFFERENT GPU
+        part = lib.create_partition(PartitionSpec(g0.uuid, CPX, 0))
+        assert part.compute_mode == CPX
+
+    def test_busy_pids_listing(self, tree, lib):
+        tree.add_kfd_process(100, 0)
+        tree.add_kfd_process(200, 0)
+        tree.add_kfd_process(300, 1)
+        b = lib.backend
+        g = lib.gpus()[0]
+        assert b.gpu_busy_pids(g.minor) == [100, 200]
+
+    def test_unreadable_proc_tree_is_not_busy(self, tree, lib):
+        # no proc dir at all -> [] (kernel EBUSY remains the backstop)
+        g = lib.gpus()[0]
+        assert lib.backend.gpu_busy_pids(g.minor) == []
+
+
+class TestPartitionWriteFallbacks:
+    """Fault-injection matrix for the three-stage partition write path
+    (VERDICT round-1 item 4): sysfs EROFS -> amdsmi lib -> amd-smi CLI."""
+
+    def _backend(self, tree):
+        # plain SysfsBackend over the mock tree (NOT MockSysfsBackend), so
+        # the real fallback chain runs
+        from k8s_dra_driver_gpu_amd.device.sysfs import SysfsBackend
+
+        return SysfsBackend(sysfs_root=tree.sysfs_root, dev_root=tree.dev_root)
+
+    def test_sysfs_write_success_short_circuits(self, tree):
+        b = self._backend(tree)
+        called = {"amdsmi": 0}
+        b._amdsmi_set_compute_partition = lambda m, mode: called.__setitem__("amdsmi", 1)
+        b.set_compute_partition(0, "CPX")
+        assert called["amdsmi"] == 0
+        assert b.get_compute_partition(0) == "CPX"
+
+    def test_erofs_falls_back_to_amdsmi(self, tree, monkeypatch):
+        import errno
+
+        b = self._backend(tree)
+        monkeypatch.setattr(
+            b, "_write_sysfs_partition",
+            lambda p, m: (_ for _ in ()).throw(OSError(errno.EROFS, "read-only fs")),
+        )
+        calls = []
+        monkeypatch.setattr(
+            b, "_amdsmi_set_compute_partition", lambda minor, mode: calls.append(mode) or True
+        )
+        b.set_compute_partition(0, "CPX")
+        assert calls == ["CPX"]
+
+    def test_amdsmi_failure_falls_back_to_cli(self, tree, monkeypatch):
+        import errno
+
+        b = self._backend(tree)
+        monkeypatch.setattr(
+            b, "_write_sysfs_partition",
+            lambda p, m: (_ for _ in ()).throw(OSError(errno.EROFS, "read-only fs")),
+        )
+        monkeypatch.setattr(b, "_amdsmi_set_compute_partition", lambda minor, mode: False)
+        calls = []
+        monkeypatch.setattr(
+            b, "_amdsmi_cli_set_compute_partition", lambda minor, mode: calls.append(mode) or True
+        )
+        b.set_compute_partition(0, "CPX")
+        assert calls == ["CPX"]
+
+    def test_all_arms_fail_raises(self, tree, monkeypatch):
+        import errno
+
+        from k8s_dra_driver_gpu_amd.device.sysfs import SysfsError
+
+        b = self._backend(tree)
+        monkeypatch.setattr(
+            b, "_write_sysfs_partition",
+            lambda p, m: (_ for _ in ()).throw(OSError(errno.EROFS, "read-only fs")),
+        )
+        monkeypatch.setattr(b, "_amdsmi_set_compute_partition", lambda minor, mode: False)
+        monkeypatch.setattr(b, "_amdsmi_cli_set_compute_partition", lambda minor, mode: False)
+        with pytest.raises(SysfsError, match="failed on card0"):
+            b.set_compute_partition(0, "CPX")
+
+    def test_ebusy_is_terminal_no_fallback(self, tree, monkeypatch):
+        """EBUSY means 'quiesce first': amdsmi/CLI would hit the same wall,
+        so the error surfaces immediately with the quiesce hint."""
+        import errno
+
+        from k8s_dra_driver_gpu_amd.device.sysfs import SysfsError
+
+        b = self._backend(tree)
+        monkeypatch.setattr(
+            b, "_write_sysfs_partition",
+            lambda p, m: (_ for _ in ()).throw(OSError(errno.EBUSY, "device busy")),
+        )
+        calls = []
+        monkeypatch.setattr(
+            b, "_amdsmi_set_compute_partition", lambda minor, mode: calls.append(1) or True
+        )
+        with pytest.raises(SysfsError, match="busy"):
+            b.set_compute_partition(0, "CPX")
+        assert calls == []
