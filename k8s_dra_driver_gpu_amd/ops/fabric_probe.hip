@@ -119,12 +119,16 @@ hbm_read_chunk_kernel(const float4v* __restrict__ src, float* __restrict__ sink,
     if (r == -1.0f) sink[0] = r;
 }
 
-// Deliberately dereference a null device pointer: generates a GPU VM page
-// fault, the KFD event the health monitor consumes as its XID-13 analog
-// (fault-injection only — used by the end-to-end health test).
+// Deliberately dereference an unmapped device address: generates a GPU VM
+// page fault, the KFD event the health monitor consumes as its XID-13
+// analog (fault-injection only — used by the end-to-end health test).
+// Reads AND writes a wild address so neither XNACK retry nor load
+// elimination can swallow the fault.
 extern "C" __global__ void vmfault_kernel(float* __restrict__ sink) {
-    const float* bad = nullptr;
-    sink[0] = bad[threadIdx.x];
+    volatile float* bad = (volatile float*)(0xdeadbeef000ull);
+    float v = bad[threadIdx.x];
+    bad[threadIdx.x] = v + 1.0f;
+    sink[0] = v;
 }
 
 extern "C" __global__ void __launch_bounds__(PROBE_BLOCK)

@@ -100,18 +100,22 @@ int fp_device_count(void) {
 
 // Trigger a GPU VM page fault on `dev` (fault injection for the health
 // monitor's end-to-end test; the process absorbs the resulting HIP error).
-// Returns 0 when the fault was provoked (sync reports an error), -1 if the
-// launch machinery itself failed.
+// Returns the hipDeviceSynchronize() error code so callers can tell whether
+// the device actually reported the fault (0 = no error surfaced, which
+// means the fault did NOT happen), or -1 if the launch machinery failed.
 int fp_trigger_vmfault(int dev) {
     if (hipSetDevice(dev) != hipSuccess) return -1;
     float* sink = nullptr;
     if (hipMalloc(&sink, sizeof(float)) != hipSuccess) return -1;
     hipLaunchKernelGGL(vmfault_kernel, dim3(1), dim3(64), 0, 0, sink);
+    hipError_t launch = hipGetLastError();
+    if (launch != hipSuccess) {
+        hipFree(sink);
+        return -1;
+    }
     hipError_t sync = hipDeviceSynchronize();
     hipFree(sink);
-    // a VM fault surfaces as an error from synchronize (or marks the
-    // context); either way the fault was raised on the device
-    return sync != hipSuccess ? 0 : 0;
+    return (int)sync;
 }
 
 // ---------------------------------------------------------------------------

@@ -104,7 +104,14 @@ class SysfsRasSource:
 
 
 class AmdSmiEventSource:
-    """amdsmi kernel event notifications (the NVML EventSet analog)."""
+    """amdsmi kernel event notifications (the NVML EventSet analog).
+
+    Wire shape verified on a real MI355X (gpurun_out/r2s3/diag.txt):
+    ``amdsmi_get_gpu_event_notification`` returns
+    ``{'num_elem': N, 'data': [{'processor_handle': int, 'event': int,
+    'message': str}]}`` — the event is the ``AmdSmiEvtNotificationType``
+    integer value and the device is identified by handle, not uuid.
+    """
 
     _KIND_MAP = {
         "VMFAULT": EVENT_VM_FAULT,
@@ -116,7 +123,7 @@ class AmdSmiEventSource:
     def __init__(self, devicelib: DeviceLib):
         self.devicelib = devicelib
         self._handles = None
-        self._uuid_by_bdf: Dict[str, str] = {}
+        self._uuid_by_handle: Dict[int, str] = {}
 
     def _ensure_init(self) -> bool:
         if self._handles is not None:
@@ -135,33 +142,65 @@ class AmdSmiEventSource:
                     if evt is not None:
                         mask |= 1 << (int(evt) - 1)
                 amdsmi.amdsmi_set_gpu_event_notification_mask(h, mask)
+                # handle -> our GPU uuid, via the PCI address
+                try:
+                    bdf = str(amdsmi.amdsmi_get_gpu_device_bdf(h)).lower()
+                    for g in self.devicelib.gpus():
+                        if bdf.endswith(g.pci_bus_id) or g.pci_bus_id.endswith(bdf):
+                            self._uuid_by_handle[self._handle_key(h)] = g.uuid
+                            break
+                except Exception:
+                    logger.debug("bdf mapping for event handle failed", exc_info=True)
             return True
         except Exception:
             logger.debug("amdsmi event source unavailable", exc_info=True)
             self._handles = None
             return False
 
+    @staticmethod
+    def _handle_key(h) -> int:
+        # amdsmi handles are ctypes pointers; events carry the raw address
+        try:
+            import ctypes
+
+            return ctypes.cast(h, ctypes.c_void_p).value or 0
+        except Exception:
+            return id(h)
+
+    def _event_name(self, value) -> str:
+        try:
+            return self._amdsmi.AmdSmiEvtNotificationType(int(value)).name
+        except Exception:
+            return str(value)
+
     def poll(self, timeout_ms: int = 1000) -> List[HealthEvent]:
         if not self._ensure_init():
             return []
         events: List[HealthEvent] = []
         try:
-            for h in self._handles or []:
-                try:
-                    raw = self._amdsmi.amdsmi_get_gpu_event_notification(timeout_ms)
-                except Exception:
-                    raw = []
-                for ev in raw or []:
-                    name = str(ev.get("event", ""))
-                    kind = self._KIND_MAP.get(name.rsplit(".", 1)[-1], name)
-                    events.append(
-                        HealthEvent(
-                            device_uuid=str(ev.get("uuid", "")) or "unknown",
-                            kind=kind,
-                            message=str(ev.get("message", "")),
-                        )
+            try:
+                raw = self._amdsmi.amdsmi_get_gpu_event_notification(timeout_ms)
+            except Exception:
+                raw = None  # AMDSMI_STATUS_NO_DATA on an empty poll
+            entries = raw.get("data", []) if isinstance(raw, dict) else (raw or [])
+            for ev in entries:
+                if not isinstance(ev, dict):
+                    continue
+                name = self._event_name(ev.get("event", ""))
+                kind = self._KIND_MAP.get(name)
+                if kind is None:
+                    continue  # PROCESS_START/END, migrations: not health events
+                uuid = self._uuid_by_handle.get(int(ev.get("processor_handle", 0) or 0), "")
+                if not uuid:
+                    gpus = self.devicelib.gpus()
+                    uuid = gpus[0].uuid if len(gpus) == 1 else "unknown"
+                events.append(
+                    HealthEvent(
+                        device_uuid=uuid,
+                        kind=kind,
+                        message=str(ev.get("message", "")),
                     )
-                break  # notifications are global per process
+                )
         except Exception:
             logger.exception("amdsmi event poll failed")
         return events
