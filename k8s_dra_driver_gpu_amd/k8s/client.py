@@ -87,8 +87,20 @@ class Client:
         raise NotImplementedError
 
     def watch(self, resource: str, namespace: Optional[str] = None,
-              selector: Optional[Dict[str, str]] = None) -> Watch:
+              selector: Optional[Dict[str, str]] = None,
+              resource_version: Optional[str] = None,
+              allow_bookmarks: bool = False) -> Watch:
+        """Watch; with ``resource_version`` the stream resumes from that RV
+        (410 Gone when too old); ``allow_bookmarks`` requests periodic
+        BOOKMARK events on idle streams."""
         raise NotImplementedError
+
+    def list_with_rv(self, resource: str, namespace: Optional[str] = None,
+                     selector: Optional[Dict[str, str]] = None):
+        """(items, collection resourceVersion). Default shim returns rv=""
+        for implementations without watch-cache semantics; the informer
+        falls back to legacy replay mode then."""
+        return self.list(resource, namespace, selector), ""
 
     def server_version(self) -> tuple:
         """(major, minor) of the API server; used to key feature availability
@@ -143,6 +155,10 @@ class Client:
 
 
 class FakeClient(Client):
+    # idle-bookmark cadence for allow_bookmarks watches (client-go asks the
+    # apiserver for bookmarks roughly every few seconds to minutes)
+    bookmark_interval = 5.0
+
     def __init__(self, server: Optional[FakeApiServer] = None):
         self.server = server or FakeApiServer()
 
@@ -164,8 +180,16 @@ class FakeClient(Client):
     def delete(self, resource, name, namespace=""):
         return self.server.delete(resource, name, namespace)
 
-    def watch(self, resource, namespace=None, selector=None):
-        return self.server.watch(resource, namespace, selector)
+    def watch(self, resource, namespace=None, selector=None,
+              resource_version=None, allow_bookmarks=False):
+        return self.server.watch(
+            resource, namespace, selector,
+            resource_version=resource_version,
+            bookmark_interval=self.bookmark_interval if allow_bookmarks else None,
+        )
+
+    def list_with_rv(self, resource, namespace=None, selector=None):
+        return self.server.list_with_rv(resource, namespace, selector)
 
     def server_version(self):
         return getattr(self.server, "version", (1, 33))
@@ -292,6 +316,15 @@ class HttpClient(Client):
         data = self._get_with_retry(self._path(resource, namespace), params=params)
         return data.get("items", [])
 
+    def list_with_rv(self, resource, namespace=None, selector=None):
+        self._throttle()
+        params = {}
+        if selector:
+            params["labelSelector"] = ",".join(f"{k}={v}" for k, v in selector.items())
+        data = self._get_with_retry(self._path(resource, namespace), params=params)
+        rv = ((data.get("metadata") or {}).get("resourceVersion")) or ""
+        return data.get("items", []), rv
+
     def update(self, resource, obj):
         self._throttle()
         md = obj.get("metadata") or {}
@@ -324,8 +357,11 @@ class HttpClient(Client):
         except Exception:
             return (1, 33)
 
-    def watch(self, resource, namespace=None, selector=None):
-        """Streamed watch; returns a Watch-like iterator thread."""
+    def watch(self, resource, namespace=None, selector=None,
+              resource_version=None, allow_bookmarks=False):
+        """Streamed watch; returns a Watch-like iterator thread. A server
+        410 (resourceVersion too old) surfaces as an ERROR WatchEvent whose
+        object carries code=410, which the informer turns into a relist."""
         from .fakeserver import WatchEvent
 
         w = Watch.__new__(Watch)
@@ -335,6 +371,8 @@ class HttpClient(Client):
         w._stopped = False
         w._server = None
         w._resource = resource
+        w._bookmark_interval = None
+        w._last_emit = 0.0
 
         def run():
             # bounded watch (client-go style): ask the server to close the
@@ -345,6 +383,10 @@ class HttpClient(Client):
 
             wt = os.environ.get("AMDDRA_WATCH_TIMEOUT", "300")
             params = {"watch": "true", "timeoutSeconds": wt}
+            if resource_version not in (None, ""):
+                params["resourceVersion"] = str(resource_version)
+            if allow_bookmarks:
+                params["allowWatchBookmarks"] = "true"
             if selector:
                 params["labelSelector"] = ",".join(f"{k}={v}" for k, v in selector.items())
             try:
@@ -353,6 +395,12 @@ class HttpClient(Client):
                     timeout=httpx.Timeout(connect=30.0, read=float(wt) + 30.0,
                                           write=30.0, pool=30.0),
                 ) as r:
+                    if r.status_code == 410:
+                        w._q.put(WatchEvent("ERROR", {"code": 410}))
+                        return
+                    if r.status_code >= 400:
+                        w._q.put(WatchEvent("ERROR", {"code": r.status_code}))
+                        return
                     for line in r.iter_lines():
                         if w._stopped:
                             return

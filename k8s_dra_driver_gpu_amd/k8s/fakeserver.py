@@ -7,8 +7,11 @@ machinery the controllers depend on:
 
 * resourceVersion bumping on every mutation, optimistic-concurrency checks
   on update,
-* list + watch (ADDED/MODIFIED/DELETED events, bookmark-free) with
-  label-selector filtering,
+* list + watch with label-selector filtering and client-go watch-cache
+  semantics: list returns a collection resourceVersion, ``watch`` resumes
+  from a given resourceVersion out of a bounded event history (a watch
+  older than the retained window gets **410 Gone**, forcing a relist), and
+  idle streams emit BOOKMARK events so clients' last-seen RV advances,
 * finalizer semantics: DELETE on an object with finalizers sets
   ``deletionTimestamp``; the object disappears when the last finalizer is
   removed,
@@ -50,6 +53,13 @@ class AlreadyExists(ApiError):
         super().__init__(409, message)
 
 
+class Gone(ApiError):
+    """410: the requested resourceVersion fell out of the watch cache."""
+
+    def __init__(self, message: str):
+        super().__init__(410, message)
+
+
 def _matches_selector(obj: Dict[str, Any], selector: Optional[Dict[str, str]]) -> bool:
     if not selector:
         return True
@@ -68,13 +78,19 @@ class WatchEvent:
 
 class Watch:
     def __init__(self, server: "FakeApiServer", resource: str, namespace: Optional[str],
-                 selector: Optional[Dict[str, str]]):
+                 selector: Optional[Dict[str, str]],
+                 bookmark_interval: Optional[float] = None):
         self._q: "queue.Queue[Optional[WatchEvent]]" = queue.Queue()
         self._server = server
         self._resource = resource
         self._namespace = namespace
         self._selector = selector
         self._stopped = False
+        # client-go-style watch bookmarks: when idle for this long, emit a
+        # BOOKMARK event carrying only metadata.resourceVersion so the
+        # client's last-seen RV advances without object traffic
+        self._bookmark_interval = bookmark_interval
+        self._last_emit = time.monotonic()
 
     def _deliver(self, event: WatchEvent) -> None:
         ns = (event.object.get("metadata") or {}).get("namespace", "")
@@ -84,6 +100,13 @@ class Watch:
             return
         self._q.put(event)
 
+    def _bookmark(self) -> WatchEvent:
+        self._last_emit = time.monotonic()
+        return WatchEvent(
+            "BOOKMARK",
+            {"metadata": {"resourceVersion": self._server.current_rv()}},
+        )
+
     def stop(self) -> None:
         self._stopped = True
         self._q.put(None)
@@ -91,27 +114,52 @@ class Watch:
 
     def __iter__(self) -> Iterator[WatchEvent]:
         while True:
-            ev = self._q.get()
+            if self._bookmark_interval is None:
+                ev = self._q.get()
+            else:
+                try:
+                    ev = self._q.get(timeout=self._bookmark_interval)
+                except queue.Empty:
+                    yield self._bookmark()
+                    continue
             if ev is None:
                 return
+            self._last_emit = time.monotonic()
             yield ev
 
     def next(self, timeout: float = 1.0) -> Optional[WatchEvent]:
         try:
-            return self._q.get(timeout=timeout)
+            ev = self._q.get(timeout=timeout)
+            if ev is not None:
+                self._last_emit = time.monotonic()
+            return ev
         except queue.Empty:
+            if (self._bookmark_interval is not None
+                    and time.monotonic() - self._last_emit >= self._bookmark_interval):
+                return self._bookmark()
             return None
 
 
 class FakeApiServer:
-    def __init__(self):
+    def __init__(self, history_limit: int = 1024):
         self._lock = threading.RLock()
         self._rv = 0
         # store[resource][(namespace, name)] = obj
         self._store: Dict[str, Dict[Tuple[str, str], Dict[str, Any]]] = {}
         self._watches: Dict[str, List[Watch]] = {}
+        # bounded per-resource event history — the watch cache that lets a
+        # client resume ``watch(resourceVersion=rv)``; events older than the
+        # window are evicted and such resumes get 410 Gone (client-go
+        # "too old resource version")
+        self._history_limit = history_limit
+        self._history: Dict[str, List[Tuple[int, str, Dict[str, Any]]]] = {}
+        self._evicted_rv: Dict[str, int] = {}
 
     # -- helpers ----------------------------------------------------------
+
+    def current_rv(self) -> str:
+        with self._lock:
+            return str(self._rv)
 
     def _bump(self, obj: Dict[str, Any]) -> None:
         self._rv += 1
@@ -122,6 +170,13 @@ class FakeApiServer:
         return (md.get("namespace", ""), md.get("name", ""))
 
     def _notify(self, resource: str, type_: str, obj: Dict[str, Any]) -> None:
+        hist = self._history.setdefault(resource, [])
+        hist.append((int(obj["metadata"]["resourceVersion"]), type_, copy.deepcopy(obj)))
+        while len(hist) > self._history_limit:
+            evicted = hist.pop(0)
+            self._evicted_rv[resource] = max(
+                self._evicted_rv.get(resource, 0), evicted[0]
+            )
         for w in list(self._watches.get(resource, [])):
             w._deliver(WatchEvent(type_, copy.deepcopy(obj)))
 
@@ -246,19 +301,57 @@ class FakeApiServer:
             self._bump(obj)
             self._notify(resource, "DELETED", obj)
 
+    def list_with_rv(
+        self,
+        resource: str,
+        namespace: Optional[str] = None,
+        selector: Optional[Dict[str, str]] = None,
+    ) -> Tuple[List[Dict[str, Any]], str]:
+        """list + the collection resourceVersion a subsequent watch resumes
+        from (k8s ListMeta.resourceVersion semantics)."""
+        with self._lock:
+            return self.list(resource, namespace, selector), str(self._rv)
+
     def watch(
         self,
         resource: str,
         namespace: Optional[str] = None,
         selector: Optional[Dict[str, str]] = None,
         send_initial: bool = True,
+        resource_version: Optional[str] = None,
+        bookmark_interval: Optional[float] = None,
     ) -> Watch:
+        """Open a watch stream.
+
+        * ``resource_version=None`` (legacy mode): current objects are
+          replayed as ADDED when ``send_initial``.
+        * ``resource_version="<rv>"``: resume — replay retained history
+          events with rv strictly greater, then go live. Raises ``Gone``
+          (410) when the requested rv precedes the retained window, exactly
+          like a real apiserver's watch cache.
+        """
         with self._lock:
-            w = Watch(self, resource, namespace, selector)
-            self._watches.setdefault(resource, []).append(w)
-            if send_initial:
-                for obj in self.list(resource, namespace, selector):
-                    w._q.put(WatchEvent("ADDED", obj))
+            w = Watch(self, resource, namespace, selector,
+                      bookmark_interval=bookmark_interval)
+            if resource_version not in (None, ""):
+                try:
+                    rv = int(resource_version)
+                except ValueError:
+                    raise ApiError(400, f"invalid resourceVersion {resource_version!r}")
+                if rv < self._evicted_rv.get(resource, 0):
+                    raise Gone(
+                        f"too old resource version: {rv} "
+                        f"(oldest retained: {self._evicted_rv.get(resource, 0) + 1})"
+                    )
+                self._watches.setdefault(resource, []).append(w)
+                for ev_rv, type_, obj in self._history.get(resource, []):
+                    if ev_rv > rv:
+                        w._deliver(WatchEvent(type_, copy.deepcopy(obj)))
+            else:
+                self._watches.setdefault(resource, []).append(w)
+                if send_initial:
+                    for obj in self.list(resource, namespace, selector):
+                        w._q.put(WatchEvent("ADDED", obj))
             return w
 
 

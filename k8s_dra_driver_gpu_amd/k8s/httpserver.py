@@ -21,7 +21,7 @@ from typing import Any, Dict, Optional, Tuple
 from urllib.parse import parse_qs, urlparse
 
 from .client import RESOURCE_INFO
-from .fakeserver import ApiError, FakeApiServer, NotFound
+from .fakeserver import ApiError, FakeApiServer, Gone, NotFound
 
 logger = logging.getLogger("amddra.httpserver")
 
@@ -105,18 +105,35 @@ class _Handler(BaseHTTPRequestHandler):
                 return
             if qs.get("watch", ["false"])[0] == "true":
                 tmo = float(qs.get("timeoutSeconds", ["0"])[0] or 0)
+                rv = qs.get("resourceVersion", [None])[0]
+                bookmarks = qs.get("allowWatchBookmarks", ["false"])[0] == "true"
                 self._stream_watch(resource, namespace or None,
-                                   _parse_selector(qs), timeout_s=tmo)
+                                   _parse_selector(qs), timeout_s=tmo,
+                                   resource_version=rv,
+                                   allow_bookmarks=bookmarks)
                 return
-            items = self.api.list(resource, namespace or None, _parse_selector(qs))
-            self._send_json(200, {"kind": "List", "items": items})
+            items, rv = self.api.list_with_rv(
+                resource, namespace or None, _parse_selector(qs))
+            self._send_json(200, {"kind": "List",
+                                  "metadata": {"resourceVersion": rv},
+                                  "items": items})
         except Exception as e:  # noqa: BLE001
             self._error(e)
 
-    def _stream_watch(self, resource, namespace, selector, timeout_s=0.0):
+    def _stream_watch(self, resource, namespace, selector, timeout_s=0.0,
+                      resource_version=None, allow_bookmarks=False):
         import time as _time
 
-        watch = self.api.watch(resource, namespace, selector)
+        try:
+            watch = self.api.watch(
+                resource, namespace, selector,
+                resource_version=resource_version,
+                bookmark_interval=5.0 if allow_bookmarks else None,
+            )
+        except Gone as e:
+            # 410: the client must relist (client-go reflector semantics)
+            self._send_json(410, {"kind": "Status", "code": 410, "message": str(e)})
+            return
         self.send_response(200)
         self.send_header("Content-Type", "application/json")
         self.send_header("Transfer-Encoding", "chunked")

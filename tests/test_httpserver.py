@@ -454,3 +454,71 @@ class TestExtendedResourceOverHttp:
         claim = client.get("resourceclaims", ercs["resourceClaimName"], "default")
         res = claim["status"]["allocation"]["devices"]["results"]
         assert res[0]["device"] == "gpu-0"
+
+
+class TestHttpWatchCache:
+    """resourceVersion resume, 410 Gone and bookmarks over the HTTP wire."""
+
+    def _pod(self, client, name):
+        return client.create("pods", {"apiVersion": "v1", "kind": "Pod",
+                                      "metadata": {"name": name,
+                                                   "namespace": "default"}})
+
+    def test_list_with_rv_over_http(self, client):
+        self._pod(client, "a")
+        items, rv = client.list_with_rv("pods")
+        assert len(items) == 1 and rv and int(rv) >= 1
+
+    def test_watch_resume_over_http(self, client):
+        self._pod(client, "a")
+        _, rv = client.list_with_rv("pods")
+        self._pod(client, "b")
+        w = client.watch("pods", resource_version=rv)
+        ev = w.next(timeout=5.0)
+        assert ev is not None and ev.type == "ADDED"
+        assert ev.object["metadata"]["name"] == "b"
+        w.stop()
+
+    def test_410_over_http_surfaces_as_error_event(self):
+        from k8s_dra_driver_gpu_amd.k8s.fakeserver import FakeApiServer
+
+        srv = MiniApiServer(api=FakeApiServer(history_limit=2))
+        srv.start()
+        try:
+            client = HttpClient(base_url=f"http://127.0.0.1:{srv.port}")
+            for i in range(6):
+                self._pod(client, f"p{i}")
+            w = client.watch("pods", resource_version="1")
+            ev = w.next(timeout=5.0)
+            assert ev is not None and ev.type == "ERROR"
+            assert ev.object.get("code") == 410
+            w.stop()
+        finally:
+            srv.stop()
+
+    def test_informer_over_http_survives_410(self):
+        from k8s_dra_driver_gpu_amd.k8s.fakeserver import FakeApiServer
+
+        srv = MiniApiServer(api=FakeApiServer(history_limit=2))
+        srv.start()
+        try:
+            client = HttpClient(base_url=f"http://127.0.0.1:{srv.port}")
+            self._pod(client, "victim")
+            inf = Informer(client, "pods").start()
+            assert inf.wait_for_sync(10.0)
+            inf._watch.stop()
+            client.delete("pods", "victim", "default")
+            for i in range(6):
+                self._pod(client, f"churn{i}")
+
+            def converged():
+                names = {o["metadata"]["name"] for o in inf.items()}
+                return "victim" not in names and "churn5" in names
+
+            deadline = time.monotonic() + 15
+            while time.monotonic() < deadline and not converged():
+                time.sleep(0.1)
+            assert converged(), [o["metadata"]["name"] for o in inf.items()]
+            inf.stop()
+        finally:
+            srv.stop()

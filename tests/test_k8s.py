@@ -473,3 +473,139 @@ class TestSchedulerMissedDelete:
         assert sched.schedule_pending() == 1
         got = c.get("resourceclaims", "b", "default")
         assert got["status"]["allocation"]["devices"]["results"][0]["device"] == "gpu-0"
+
+
+class TestWatchCacheSemantics:
+    """client-go watch-cache model (VERDICT round-1 item 9):
+    resourceVersion resume, 410 Gone, BOOKMARK events."""
+
+    def _mk(self, s, name, ns="default"):
+        return s.create("pods", {"apiVersion": "v1", "kind": "Pod",
+                                 "metadata": {"name": name, "namespace": ns}})
+
+    def test_list_with_rv_and_resume(self):
+        from k8s_dra_driver_gpu_amd.k8s.fakeserver import FakeApiServer
+
+        s = FakeApiServer()
+        self._mk(s, "a")
+        items, rv = s.list_with_rv("pods")
+        assert len(items) == 1 and rv == items[0]["metadata"]["resourceVersion"]
+        self._mk(s, "b")
+        s.delete("pods", "a", "default")
+        w = s.watch("pods", resource_version=rv)
+        evs = [w.next(timeout=0.5) for _ in range(2)]
+        assert [(e.type, e.object["metadata"]["name"]) for e in evs] == [
+            ("ADDED", "b"), ("DELETED", "a")]
+        assert w.next(timeout=0.1) is None  # nothing replayed from before rv
+        w.stop()
+
+    def test_too_old_rv_gets_410_gone(self):
+        from k8s_dra_driver_gpu_amd.k8s.fakeserver import FakeApiServer, Gone
+
+        s = FakeApiServer(history_limit=2)
+        for i in range(6):
+            self._mk(s, f"p{i}")
+        with pytest.raises(Gone):
+            s.watch("pods", resource_version="1")
+        # a recent rv still resumes fine
+        _, rv = s.list_with_rv("pods")
+        w = s.watch("pods", resource_version=rv)
+        assert w.next(timeout=0.1) is None
+        w.stop()
+
+    def test_bookmark_on_idle_stream(self):
+        from k8s_dra_driver_gpu_amd.k8s.fakeserver import FakeApiServer
+
+        s = FakeApiServer()
+        self._mk(s, "a")
+        _, rv = s.list_with_rv("pods")
+        w = s.watch("pods", resource_version=rv, bookmark_interval=0.1)
+        time.sleep(0.15)
+        ev = w.next(timeout=0.3)
+        assert ev is not None and ev.type == "BOOKMARK"
+        assert ev.object["metadata"]["resourceVersion"] == rv
+        w.stop()
+
+    def test_informer_survives_410_without_missing_deletes(self):
+        """Forced 410: the informer is disconnected while the event window
+        is evicted (including a DELETED it never saw); the relist driven by
+        410 reconciles the cache — no controller-level resync crutch."""
+        from k8s_dra_driver_gpu_amd.k8s.client import FakeClient
+        from k8s_dra_driver_gpu_amd.k8s.fakeserver import FakeApiServer
+        from k8s_dra_driver_gpu_amd.k8s.informer import Informer
+
+        s = FakeApiServer(history_limit=2)
+        c = FakeClient(s)
+        self._mk(s, "victim")
+        events = []
+        inf = Informer(c, "pods").start()
+        assert inf.wait_for_sync(5.0)
+        inf.add_handler(lambda t, o: events.append((t, o["metadata"]["name"])))
+        # simulate a network partition: kill the informer's live stream
+        inf._watch.stop()
+        # while disconnected: the victim is deleted and enough churn evicts
+        # the event from the 2-entry history window
+        s.delete("pods", "victim", "default")
+        for i in range(6):
+            self._mk(s, f"churn{i}")
+
+        def converged():
+            names = {o["metadata"]["name"] for o in inf.items()}
+            return "victim" not in names and "churn5" in names
+
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline and not converged():
+            time.sleep(0.05)
+        assert converged(), [o["metadata"]["name"] for o in inf.items()]
+        assert ("DELETED", "victim") in events
+        inf.stop()
+
+    def test_informer_rewatches_without_relist_when_rv_retained(self):
+        """A plain stream break with the rv still in the window must NOT
+        trigger a relist — the re-watch resumes from last_resource_version."""
+        from k8s_dra_driver_gpu_amd.k8s.client import FakeClient
+        from k8s_dra_driver_gpu_amd.k8s.fakeserver import FakeApiServer
+        from k8s_dra_driver_gpu_amd.k8s.informer import Informer
+
+        s = FakeApiServer()
+        c = FakeClient(s)
+        lists = []
+        real = c.list_with_rv
+        c.list_with_rv = lambda *a, **k: (lists.append(1), real(*a, **k))[1]
+        self._mk(s, "a")
+        inf = Informer(c, "pods").start()
+        assert inf.wait_for_sync(5.0)
+        assert len(lists) == 1
+        inf._watch.stop()  # stream break, rv still retained
+        self._mk(s, "b")
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline and inf.get("default/b") is None:
+            time.sleep(0.05)
+        assert inf.get("default/b") is not None
+        assert len(lists) == 1, "re-watch must resume from rv, not relist"
+        inf.stop()
+
+    def test_bookmark_advances_informer_rv(self):
+        """Idle informer's last-seen rv advances via bookmarks, so a later
+        re-watch does not resume from a stale (evictable) rv."""
+        from k8s_dra_driver_gpu_amd.k8s.client import FakeClient
+        from k8s_dra_driver_gpu_amd.k8s.fakeserver import FakeApiServer
+        from k8s_dra_driver_gpu_amd.k8s.informer import Informer
+
+        s = FakeApiServer()
+        c = FakeClient(s)
+        c.bookmark_interval = 0.1
+        self._mk(s, "a")
+        inf = Informer(c, "pods").start()
+        assert inf.wait_for_sync(5.0)
+        rv0 = int(inf.last_resource_version)
+        # churn on a DIFFERENT resource bumps the global rv; the pods
+        # watch sees no object events, only bookmarks
+        for i in range(5):
+            s.create("nodes", {"apiVersion": "v1", "kind": "Node",
+                               "metadata": {"name": f"n{i}"}})
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline and int(inf.last_resource_version) <= rv0:
+            time.sleep(0.05)
+        assert int(inf.last_resource_version) > rv0
+        inf.stop()
