@@ -506,6 +506,9 @@ class TestHttpWatchCache:
             self._pod(client, "victim")
             inf = Informer(client, "pods").start()
             assert inf.wait_for_sync(10.0)
+            deadline = time.monotonic() + 5
+            while time.monotonic() < deadline and inf._watch is None:
+                time.sleep(0.02)
             inf._watch.stop()
             client.delete("pods", "victim", "default")
             for i in range(6):

@@ -542,6 +542,9 @@ class TestWatchCacheSemantics:
         assert inf.wait_for_sync(5.0)
         inf.add_handler(lambda t, o: events.append((t, o["metadata"]["name"])))
         # simulate a network partition: kill the informer's live stream
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline and inf._watch is None:
+            time.sleep(0.02)
         inf._watch.stop()
         # while disconnected: the victim is deleted and enough churn evicts
         # the event from the 2-entry history window
@@ -576,6 +579,9 @@ class TestWatchCacheSemantics:
         inf = Informer(c, "pods").start()
         assert inf.wait_for_sync(5.0)
         assert len(lists) == 1
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline and inf._watch is None:
+            time.sleep(0.02)
         inf._watch.stop()  # stream break, rv still retained
         self._mk(s, "b")
         deadline = time.monotonic() + 10
