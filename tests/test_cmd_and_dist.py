@@ -179,20 +179,29 @@ class TestInspectCli:
 
 
 class TestHelmChart:
-    """The chart's templates keep Go-templating inside quoted scalars, so
-    every manifest is YAML-parseable as-shipped — structural lint the
-    reference gets from `helm template` in CI (tests/bats/test_basics.bats)."""
+    """Chart rendered through helmlite (the in-repo Helm-subset renderer;
+    no helm binary in the image) — the structural verification the
+    reference gets from `helm template` in CI (tests/bats/test_basics.bats),
+    plus golden checks for the templated knobs."""
 
     CHART = os.path.join(REPO, "deployments", "helm", "amd-dra-driver")
 
-    def _docs(self):
+    def _render(self, overrides=None, **kw):
+        from k8s_dra_driver_gpu_amd.utils.helmlite import render_chart
+
+        return render_chart(self.CHART, overrides or {}, **kw)
+
+    def _docs(self, overrides=None, **kw):
         import glob
 
         import yaml
 
         docs = []
-        for f in (glob.glob(os.path.join(self.CHART, "templates", "*.yaml"))
-                  + glob.glob(os.path.join(self.CHART, "crds", "*.yaml"))):
+        for text in self._render(overrides, **kw).values():
+            for d in yaml.safe_load_all(text):
+                if d:
+                    docs.append(d)
+        for f in glob.glob(os.path.join(self.CHART, "crds", "*.yaml")):
             for d in yaml.safe_load_all(open(f)):
                 if d:
                     docs.append(d)
@@ -234,9 +243,95 @@ class TestHelmChart:
                         "computedomaincliques.resource.amd.com"}
         dcs = {d["metadata"]["name"] for d in docs if d["kind"] == "DeviceClass"}
         assert {"gpu.amd.com", "partition.gpu.amd.com", "vfio.gpu.amd.com"} <= dcs
+        # default (v1beta1): extendedResourceName must NOT appear — it is a
+        # v1-only field (ref deviceclass-gpu.yaml:12-14)
         gpu_dc = next(d for d in docs if d["kind"] == "DeviceClass"
                       and d["metadata"]["name"] == "gpu.amd.com")
+        assert gpu_dc["apiVersion"] == "resource.k8s.io/v1beta1"
+        assert "extendedResourceName" not in gpu_dc["spec"]
+
+    def test_resource_api_version_templating(self):
+        """resourceApiVersion=v1: classes serve under resource.k8s.io/v1 and
+        gpu.amd.com gains extendedResourceName."""
+        docs = self._docs({"resourceApiVersion": "v1"})
+        dcs = [d for d in docs if d["kind"] == "DeviceClass"]
+        assert all(d["apiVersion"] == "resource.k8s.io/v1" for d in dcs)
+        gpu_dc = next(d for d in dcs if d["metadata"]["name"] == "gpu.amd.com")
         assert gpu_dc["spec"]["extendedResourceName"] == "amd.com/gpu"
+
+    def test_network_policies_gated_and_shaped(self):
+        docs = self._docs()
+        assert not any(d["kind"] == "NetworkPolicy" for d in docs)
+        docs = self._docs({
+            "controller": {"networkPolicy": {"enabled": True}},
+            "kubeletPlugin": {"networkPolicy": {"enabled": True}},
+            "webhook": {"networkPolicy": {"enabled": True}},
+        })
+        nps = {d["metadata"]["name"]: d for d in docs if d["kind"] == "NetworkPolicy"}
+        assert set(nps) == {"amd-dra-controller", "amd-dra-kubelet-plugin",
+                            "amd-dra-webhook"}
+        kp = nps["amd-dra-kubelet-plugin"]
+        ports = {p["port"] for rule in kp["spec"]["egress"] for p in rule["ports"]}
+        assert {443, 6443, 50000, 50005} <= ports  # fabricd mesh ports
+        wh = nps["amd-dra-webhook"]
+        assert "Ingress" in wh["spec"]["policyTypes"]
+
+    def test_webhook_cert_manager_mode(self):
+        docs = self._docs()  # default: cert-manager + selfsigned
+        kinds = {d["kind"] for d in docs}
+        assert {"Issuer", "Certificate"} <= kinds
+        cert = next(d for d in docs if d["kind"] == "Certificate")
+        assert cert["spec"]["secretName"] == "amd-dra-webhook-cert"
+        assert cert["spec"]["issuerRef"]["name"] == "amd-dra-webhook-issuer"
+        assert any("svc" in n for n in cert["spec"]["dnsNames"])
+        vwc = next(d for d in docs if d["kind"] == "ValidatingWebhookConfiguration")
+        assert "cert-manager.io/inject-ca-from" in vwc["metadata"]["annotations"]
+        dep = next(d for d in docs if d["kind"] == "Deployment"
+                   and d["metadata"]["name"] == "amd-dra-webhook")
+        vol = dep["spec"]["template"]["spec"]["volumes"][0]
+        assert vol["secret"]["secretName"] == "amd-dra-webhook-cert"
+
+    def test_webhook_secret_mode(self):
+        docs = self._docs({"webhook": {"tls": {
+            "mode": "secret", "secretName": "my-tls", "caBundle": "QUJD"}}})
+        kinds = {d["kind"] for d in docs}
+        assert "Issuer" not in kinds and "Certificate" not in kinds
+        vwc = next(d for d in docs if d["kind"] == "ValidatingWebhookConfiguration")
+        assert vwc["webhooks"][0]["clientConfig"]["caBundle"] == "QUJD"
+        dep = next(d for d in docs if d["kind"] == "Deployment"
+                   and d["metadata"]["name"] == "amd-dra-webhook")
+        assert dep["spec"]["template"]["spec"]["volumes"][0]["secret"]["secretName"] == "my-tls"
+
+    def test_webhook_disabled_removes_all_webhook_objects(self):
+        docs = self._docs({"webhook": {"enabled": False}})
+        for d in docs:
+            assert "webhook" not in d["metadata"]["name"], d["metadata"]["name"]
+
+    def test_validation_rejects_bad_values(self):
+        from k8s_dra_driver_gpu_amd.utils.helmlite import HelmliteError
+
+        with pytest.raises(HelmliteError, match="tls.mode"):
+            self._render({"webhook": {"tls": {"mode": "bogus"}}})
+        with pytest.raises(HelmliteError, match="resourceApiVersion"):
+            self._render({"resourceApiVersion": "v2"})
+
+    def test_standard_labels_on_all_objects(self):
+        docs = self._docs()
+        for d in docs:
+            if d["kind"] == "CustomResourceDefinition":
+                continue  # crds/ ship unrendered
+            labels = d["metadata"].get("labels") or {}
+            assert labels.get("app.kubernetes.io/name") == "amd-dra-driver", d["metadata"]
+            assert labels.get("app.kubernetes.io/managed-by") == "Helm"
+            assert "helm.sh/chart" in labels
+
+    def test_namespace_override(self):
+        docs = self._docs({"namespaceOverride": "custom-ns"})
+        namespaced = [d for d in docs if d["kind"] in
+                      ("Deployment", "DaemonSet", "Service", "ServiceAccount")]
+        assert namespaced
+        for d in namespaced:
+            assert d["metadata"]["namespace"] == "custom-ns", d["metadata"]
 
     def test_template_value_refs_resolve(self):
         """Every `.Values.x.y` referenced by a template must exist in
