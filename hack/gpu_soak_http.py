@@ -33,11 +33,11 @@ def main() -> int:
     srv = MiniApiServer()
     srv.start()
     client = HttpClient(base_url=f"http://127.0.0.1:{srv.port}", qps=10000, burst=10000)
-    chart = os.path.join(REPO, "deployments", "helm", "amd-dra-driver",
-                         "templates", "deviceclasses.yaml")
-    for doc in yaml.safe_load_all(open(chart)):
-        if doc:
-            client.create("deviceclasses", doc)
+    from k8s_dra_driver_gpu_amd.utils.helmlite import chart_deviceclasses
+
+    chart = os.path.join(REPO, "deployments", "helm", "amd-dra-driver")
+    for doc in chart_deviceclasses(chart):
+        client.create("deviceclasses", doc)
 
     work = tempfile.mkdtemp(prefix="httpsoak-")
     env = dict(os.environ)

@@ -99,11 +99,15 @@ class LocalCluster:
     # -- lifecycle ----------------------------------------------------------
 
     def start(self) -> "LocalCluster":
-        # DeviceClasses + node object
-        with open(os.path.join(CHART_DIR, "templates", "deviceclasses.yaml")) as f:
-            for doc in yaml.safe_load_all(f):
-                if doc:
-                    self.client.create("deviceclasses", doc)
+        # DeviceClasses + node object (chart rendered via helmlite; this
+        # cluster models k8s >= 1.35, so classes serve under resource.k8s.io/v1
+        # with extendedResourceName on gpu.amd.com)
+        from ..utils.helmlite import render_chart
+
+        rendered = render_chart(CHART_DIR, {"resourceApiVersion": "v1"})
+        for doc in yaml.safe_load_all(rendered["deviceclasses.yaml"]):
+            if doc:
+                self.client.create("deviceclasses", doc)
         self.client.create(
             "nodes", {"apiVersion": "v1", "kind": "Node",
                       "metadata": {"name": self.node_name}}

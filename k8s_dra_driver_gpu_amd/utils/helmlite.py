@@ -430,6 +430,17 @@ def render_chart(chart_dir: str, values_override: Optional[Dict[str, Any]] = Non
     return out
 
 
+def chart_deviceclasses(chart_dir: str,
+                        resource_api_version: str = "v1") -> List[Dict[str, Any]]:
+    """Render the chart's DeviceClass manifests (default: the v1 surface a
+    k8s >= 1.35 cluster serves, with extendedResourceName on gpu.amd.com).
+    Shared by the e2e tests, bench localcluster and soak harnesses."""
+    import yaml
+
+    rendered = render_chart(chart_dir, {"resourceApiVersion": resource_api_version})
+    return [d for d in yaml.safe_load_all(rendered["deviceclasses.yaml"]) if d]
+
+
 def render_template(src: str, values: Dict[str, Any],
                     defines_src: str = "",
                     release: Optional[Dict[str, str]] = None,

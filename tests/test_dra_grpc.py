@@ -290,11 +290,13 @@ class TestVfioAdvertisement:
         tree.setup()
         lib = DeviceLib(backend=tree.backend())
         sl = ResourceSliceGenerator(lib, node_name="n1", vfio=True).generate()[0]
+        from k8s_dra_driver_gpu_amd.utils.helmlite import chart_deviceclasses
+
         chart = os.path.join(
             os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
-            "deployments", "helm", "amd-dra-driver", "templates", "deviceclasses.yaml",
+            "deployments", "helm", "amd-dra-driver",
         )
-        dcs = {d["metadata"]["name"]: d for d in _yaml.safe_load_all(open(chart)) if d}
+        dcs = {d["metadata"]["name"]: d for d in chart_deviceclasses(chart)}
         v = next(d for d in sl["spec"]["devices"] if d["name"] == "gpu-0-vfio")
         assert device_matches_class(v, "gpu.amd.com", dcs["vfio.gpu.amd.com"])
         assert not device_matches_class(v, "gpu.amd.com", dcs["gpu.amd.com"])

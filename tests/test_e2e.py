@@ -35,10 +35,12 @@ CHART = os.path.join(
 
 
 def load_device_classes(client):
-    with open(os.path.join(CHART, "templates", "deviceclasses.yaml")) as f:
-        for doc in yaml.safe_load_all(f):
-            if doc:
-                client.create("deviceclasses", doc)
+    from k8s_dra_driver_gpu_amd.utils.helmlite import render_chart
+
+    rendered = render_chart(CHART, {"resourceApiVersion": "v1"})
+    for doc in yaml.safe_load_all(rendered["deviceclasses.yaml"]):
+        if doc:
+            client.create("deviceclasses", doc)
 
 
 def wait_for(fn, timeout=20.0, interval=0.05):

@@ -159,12 +159,11 @@ class TestMultiProcessCluster:
         from k8s_dra_driver_gpu_amd.dra import api as dra
         import yaml
 
-        chart = os.path.join(REPO, "deployments", "helm", "amd-dra-driver",
-                             "templates", "deviceclasses.yaml")
-        with open(chart) as f:
-            for doc in yaml.safe_load_all(f):
-                if doc:
-                    client.create("deviceclasses", doc)
+        from k8s_dra_driver_gpu_amd.utils.helmlite import chart_deviceclasses
+
+        chart = os.path.join(REPO, "deployments", "helm", "amd-dra-driver")
+        for doc in chart_deviceclasses(chart):
+            client.create("deviceclasses", doc)
         tree = MockTree(root=str(tmp_path / "mock"), num_gpus=1)
         tree.setup()
         env = dict(os.environ)
@@ -282,12 +281,11 @@ class TestWatchRecycleChurn:
         healed by the periodic resync within SCHED_RESYNC_INTERVAL)."""
         import yaml
 
-        chart = os.path.join(REPO, "deployments", "helm", "amd-dra-driver",
-                             "templates", "deviceclasses.yaml")
-        with open(chart) as f:
-            for doc in yaml.safe_load_all(f):
-                if doc:
-                    client.create("deviceclasses", doc)
+        from k8s_dra_driver_gpu_amd.utils.helmlite import chart_deviceclasses
+
+        chart = os.path.join(REPO, "deployments", "helm", "amd-dra-driver")
+        for doc in chart_deviceclasses(chart):
+            client.create("deviceclasses", doc)
         client.create("resourceslices", {
             "apiVersion": "resource.k8s.io/v1beta1", "kind": "ResourceSlice",
             "metadata": {"name": "wr-gpu"},
