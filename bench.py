@@ -425,7 +425,10 @@ def main() -> None:
                     "higher_is_better": True,
                     "scaling": "weak",
                     "vs_baseline": None,
-                    "dtype": "bf16",
+                    # control-plane metric: no compute dtype is involved in
+                    # the timed region (claim churn over gRPC + checkpoint
+                    # fsyncs + CDI writes)
+                    "dtype": "n/a",
                     "data": data_kind,
                     "config": {
                         "model": "dra-claim-churn",

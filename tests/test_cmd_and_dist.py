@@ -391,7 +391,7 @@ class TestBenchContract:
         assert out["higher_is_better"] is True
         assert out["scaling"] == "weak"
         assert out["vs_baseline"] is None  # no published reference number
-        assert out["dtype"] == "bf16"
+        assert out["dtype"] == "n/a"  # control-plane metric: no compute dtype
         assert "synthetic" in out["data"]
         cfg = out["config"]
         assert cfg["model"] == "dra-claim-churn"
