@@ -610,3 +610,491 @@ extern "C" __global__ void __launch_bounds__(THREADS)
 gemm_bf16_128_pipe3_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
     gemm_bf16_128_pipe3_body<32>(A, Bt, C, M, N, K);
 }
+
+// ---------------------------------------------------------------------------
+// Ladder step (round 2): register-hoisted fragments. The pipe2 inner loop
+// re-reads each B fragment from LDS for every i (16 b-reads + 4 a-reads per
+// 16 MFMAs = 1.25 ds_read_b128 per MFMA). Loading a[4] and b[4] into
+// registers once per K-substep cuts LDS reads to 8 per 16 MFMAs
+// (0.5/MFMA) at +32 VGPRs — measures whether the 16x16x32 tile is
+// ds_read-bound at depth-2 (the round-1 32x32x16 experiment said "not
+// LDS-bound" at 1 read/MFMA granularity; this probes the other direction).
+// ---------------------------------------------------------------------------
+
+template <int BK>
+__device__ __forceinline__ void gemm_bf16_128_pipe2r_body(
+    const short* __restrict__ A, const short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+    __shared__ short lds[4 * (BM * BK + BN * BK)];
+    const int HALF = BM * BK + BN * BK;
+    auto ldsA = [&](int buf) -> short* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> short* { return lds + buf * HALF + BM * BK; };
+
+    const int tiles_n = (N + BN - 1) / BN;
+    const int tile_m = blockIdx.x / tiles_n;
+    const int tile_n = blockIdx.x % tiles_n;
+    const int m0 = tile_m * BM;
+    const int n0 = tile_n * BN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid >> 1) * 64;
+    const int wc = (wid & 1) * 64;
+
+    constexpr int CHUNKS = BM * (BK / 8);
+    constexpr int PHASES = CHUNKS / THREADS;
+    constexpr int S = PHASES * 2;
+    auto stage = [&](int buf, int k0) {
+        short* la = ldsA(buf);
+        short* lb = ldsB(buf);
+#pragma unroll
+        for (int phase = 0; phase < PHASES; ++phase) {
+            int chunk = phase * THREADS + tid;
+            int r = chunk / (BK / 8);
+            int c = chunk % (BK / 8);
+            const short* ga = &A[(size_t)(m0 + r) * K + k0 + c * 8];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * THREADS + wid * WAVE) * 8),
+                16, 0, 0);
+            const short* gb = &Bt[(size_t)(n0 + r) * K + k0 + c * 8];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * THREADS + wid * WAVE) * 8),
+                16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][4] = {};
+    const int kg = (lane >> 4) * 8;
+    const int steps = K / BK;
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    if (steps > 2) stage(2, 2 * BK);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s & 3;
+        if (s + 2 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(2 * S) : "memory");
+        else if (s + 1 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 3 < steps) stage((s + 3) & 3, (s + 3) * BK);
+
+        const short* la = ldsA(buf);
+        const short* lb = ldsB(buf);
+#pragma unroll
+        for (int ks = 0; ks < BK / 32; ++ks) {
+            bf16x8 af[4], bf[4];
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                const int ar = wr + i * 16 + (lane & 15);
+                af[i] = *(const bf16x8*)&la[ar * BK + ks * 32 + kg];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int bc = wc + j * 16 + (lane & 15);
+                bf[j] = *(const bf16x8*)&lb[bc * BK + ks * 32 + kg];
+            }
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr + i * 16 + (lane >> 4) * 4 + r;
+                int col = n0 + wc + j * 16 + (lane & 15);
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(THREADS)
+gemm_bf16_128_pipe2r_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
+    gemm_bf16_128_pipe2r_body<32>(A, Bt, C, M, N, K);
+}
+
+// ---------------------------------------------------------------------------
+// Ladder step (round 2): 256x128 output tile, 512 threads = 8 waves (4x2
+// grid of 64x64 wave tiles), register-hoisted fragments, 3-buffer counted
+// pipeline. Doubles the A-reuse per staged B tile (each B tile feeds 2x the
+// MFMAs) and halves the number of workgroups (L2 tile traffic), at
+// 3 x 24 KiB = 72 KiB LDS -> 2 workgroups/CU (16 waves).
+// ---------------------------------------------------------------------------
+
+#define BM2 256
+#define THREADS2 512
+
+template <int BK>
+__device__ __forceinline__ void gemm_bf16_256_body(
+    const short* __restrict__ A, const short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+    __shared__ short lds[3 * (BM2 * BK + BN * BK)];
+    const int HALF = BM2 * BK + BN * BK;
+    auto ldsA = [&](int buf) -> short* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> short* { return lds + buf * HALF + BM2 * BK; };
+
+    const int tiles_n = (N + BN - 1) / BN;
+    const int tile_m = blockIdx.x / tiles_n;
+    const int tile_n = blockIdx.x % tiles_n;
+    const int m0 = tile_m * BM2;
+    const int n0 = tile_n * BN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;          // 8 waves: 4x2 grid of 64x64
+    const int wr = (wid >> 1) * 64;
+    const int wc = (wid & 1) * 64;
+
+    // A: BM2*(BK/8) = 1024 chunks (2 phases of 512); B: BN*(BK/8) = 512 (1)
+    constexpr int ACH = BM2 * (BK / 8);
+    constexpr int APH = ACH / THREADS2;
+    constexpr int BCH = BN * (BK / 8);
+    constexpr int BPH = BCH / THREADS2;
+    constexpr int S = APH + BPH;  // glds per thread per stage
+    auto stage = [&](int buf, int k0) {
+        short* la = ldsA(buf);
+        short* lb = ldsB(buf);
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * THREADS2 + tid;
+            int r = chunk / (BK / 8);
+            int c = chunk % (BK / 8);
+            const short* ga = &A[(size_t)(m0 + r) * K + k0 + c * 8];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * THREADS2 + wid * WAVE) * 8),
+                16, 0, 0);
+        }
+#pragma unroll
+        for (int phase = 0; phase < BPH; ++phase) {
+            int chunk = phase * THREADS2 + tid;
+            int r = chunk / (BK / 8);
+            int c = chunk % (BK / 8);
+            const short* gb = &Bt[(size_t)(n0 + r) * K + k0 + c * 8];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * THREADS2 + wid * WAVE) * 8),
+                16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][4] = {};
+    const int kg = (lane >> 4) * 8;
+    const int steps = K / BK;
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s % 3;
+        if (s + 1 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 2 < steps) stage((s + 2) % 3, (s + 2) * BK);
+
+        const short* la = ldsA(buf);
+        const short* lb = ldsB(buf);
+#pragma unroll
+        for (int ks = 0; ks < BK / 32; ++ks) {
+            bf16x8 af[4], bf[4];
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                const int ar = wr + i * 16 + (lane & 15);
+                af[i] = *(const bf16x8*)&la[ar * BK + ks * 32 + kg];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int bc = wc + j * 16 + (lane & 15);
+                bf[j] = *(const bf16x8*)&lb[bc * BK + ks * 32 + kg];
+            }
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr + i * 16 + (lane >> 4) * 4 + r;
+                int col = n0 + wc + j * 16 + (lane & 15);
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(THREADS2)
+gemm_bf16_256_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
+    gemm_bf16_256_body<32>(A, Bt, C, M, N, K);
+}
+
+// ---------------------------------------------------------------------------
+// Generic big-tile body (round-2 ladder): TWRxTWC waves of 64-lane wavefronts
+// tile a TBMxTBN output block; each wave computes AI x BJ fragments of
+// v_mfma_f32_16x16x32_bf16 with register-hoisted fragments; 3-buffer counted
+// pipeline (one tile in flight across each barrier).
+//   gemm_bf16_256 (732)   = Big<256,128, 4,2> measured 1055/1187 TF
+//   gemm_bf16_256x256     = Big<256,256, 4,2> (wave = 64x128: AI=4, BJ=8)
+// (a 512x128 variant was rejected: its B tile has fewer 16 B chunks than
+// threads, breaking the per-thread counted-vmcnt discipline; and 512-wide
+// tiles underfill the 256-CU chip at 4096^2 output)
+// ---------------------------------------------------------------------------
+
+template <int TBM, int TBN, int TWR, int TWC, int BK>
+__device__ __forceinline__ void gemm_big_body(
+    const short* __restrict__ A, const short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+    constexpr int NT = TWR * TWC * WAVE;        // threads
+    constexpr int WM = TBM / TWR;               // wave tile rows
+    constexpr int WN = TBN / TWC;               // wave tile cols
+    constexpr int AI = WM / 16;                 // a-fragments per wave
+    constexpr int BJ = WN / 16;                 // b-fragments per wave
+    __shared__ short lds[3 * (TBM * BK + TBN * BK)];
+    const int HALF = TBM * BK + TBN * BK;
+    auto ldsA = [&](int buf) -> short* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> short* { return lds + buf * HALF + TBM * BK; };
+
+    const int tiles_n = (N + TBN - 1) / TBN;
+    const int m0 = (blockIdx.x / tiles_n) * TBM;
+    const int n0 = (blockIdx.x % tiles_n) * TBN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid / TWC) * WM;
+    const int wc = (wid % TWC) * WN;
+
+    constexpr int ACH = TBM * (BK / 8);
+    constexpr int APH = ACH / NT;
+    constexpr int BCH = TBN * (BK / 8);
+    constexpr int BPH = BCH / NT;
+    static_assert(APH * NT == ACH && BPH * NT == BCH, "phase split");
+    constexpr int S = APH + BPH;
+    auto stage = [&](int buf, int k0) {
+        short* la = ldsA(buf);
+        short* lb = ldsB(buf);
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 8);
+            int c = chunk % (BK / 8);
+            const short* ga = &A[(size_t)(m0 + r) * K + k0 + c * 8];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 8),
+                16, 0, 0);
+        }
+#pragma unroll
+        for (int phase = 0; phase < BPH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 8);
+            int c = chunk % (BK / 8);
+            const short* gb = &Bt[(size_t)(n0 + r) * K + k0 + c * 8];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 8),
+                16, 0, 0);
+        }
+    };
+
+    f32x4 acc[AI][BJ] = {};
+    const int kg = (lane >> 4) * 8;
+    const int steps = K / BK;
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s % 3;
+        if (s + 1 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 2 < steps) stage((s + 2) % 3, (s + 2) * BK);
+
+        const short* la = ldsA(buf);
+        const short* lb = ldsB(buf);
+#pragma unroll
+        for (int ks = 0; ks < BK / 32; ++ks) {
+            bf16x8 af[AI], bf[BJ];
+#pragma unroll
+            for (int i = 0; i < AI; ++i) {
+                const int ar = wr + i * 16 + (lane & 15);
+                af[i] = *(const bf16x8*)&la[ar * BK + ks * 32 + kg];
+            }
+#pragma unroll
+            for (int j = 0; j < BJ; ++j) {
+                const int bc = wc + j * 16 + (lane & 15);
+                bf[j] = *(const bf16x8*)&lb[bc * BK + ks * 32 + kg];
+            }
+#pragma unroll
+            for (int i = 0; i < AI; ++i)
+#pragma unroll
+                for (int j = 0; j < BJ; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < AI; ++i)
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr + i * 16 + (lane >> 4) * 4 + r;
+                int col = n0 + wc + j * 16 + (lane & 15);
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_bf16_256x256_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
+    gemm_big_body<256, 256, 4, 2, 32>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_bf16_256_bk64_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
+    gemm_big_body<256, 128, 4, 2, 64>(A, Bt, C, M, N, K);
+}
+
+// Depth-2 variant of the generic big-tile body: 4 LDS buffers, TWO tiles in
+// flight across each barrier (pipe2-style counted vmcnt). For 256x256 this
+// is 4 x 32 KiB = 128 KiB LDS (fits the 160 KiB CU budget at 1 WG/CU).
+template <int TBM, int TBN, int TWR, int TWC, int BK>
+__device__ __forceinline__ void gemm_big2_body(
+    const short* __restrict__ A, const short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+    constexpr int NT = TWR * TWC * WAVE;
+    constexpr int WM = TBM / TWR;
+    constexpr int WN = TBN / TWC;
+    constexpr int AI = WM / 16;
+    constexpr int BJ = WN / 16;
+    __shared__ short lds[4 * (TBM * BK + TBN * BK)];
+    const int HALF = TBM * BK + TBN * BK;
+    auto ldsA = [&](int buf) -> short* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> short* { return lds + buf * HALF + TBM * BK; };
+
+    const int tiles_n = (N + TBN - 1) / TBN;
+    const int m0 = (blockIdx.x / tiles_n) * TBM;
+    const int n0 = (blockIdx.x % tiles_n) * TBN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid / TWC) * WM;
+    const int wc = (wid % TWC) * WN;
+
+    constexpr int ACH = TBM * (BK / 8);
+    constexpr int APH = ACH / NT;
+    constexpr int BCH = TBN * (BK / 8);
+    constexpr int BPH = BCH / NT;
+    static_assert(APH * NT == ACH && BPH * NT == BCH, "phase split");
+    constexpr int S = APH + BPH;
+    auto stage = [&](int buf, int k0) {
+        short* la = ldsA(buf);
+        short* lb = ldsB(buf);
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 8);
+            int c = chunk % (BK / 8);
+            const short* ga = &A[(size_t)(m0 + r) * K + k0 + c * 8];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 8),
+                16, 0, 0);
+        }
+#pragma unroll
+        for (int phase = 0; phase < BPH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 8);
+            int c = chunk % (BK / 8);
+            const short* gb = &Bt[(size_t)(n0 + r) * K + k0 + c * 8];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 8),
+                16, 0, 0);
+        }
+    };
+
+    f32x4 acc[AI][BJ] = {};
+    const int kg = (lane >> 4) * 8;
+    const int steps = K / BK;
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    if (steps > 2) stage(2, 2 * BK);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s & 3;
+        if (s + 2 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(2 * S) : "memory");
+        else if (s + 1 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 3 < steps) stage((s + 3) & 3, (s + 3) * BK);
+
+        const short* la = ldsA(buf);
+        const short* lb = ldsB(buf);
+#pragma unroll
+        for (int ks = 0; ks < BK / 32; ++ks) {
+            bf16x8 af[AI], bf[BJ];
+#pragma unroll
+            for (int i = 0; i < AI; ++i) {
+                const int ar = wr + i * 16 + (lane & 15);
+                af[i] = *(const bf16x8*)&la[ar * BK + ks * 32 + kg];
+            }
+#pragma unroll
+            for (int j = 0; j < BJ; ++j) {
+                const int bc = wc + j * 16 + (lane & 15);
+                bf[j] = *(const bf16x8*)&lb[bc * BK + ks * 32 + kg];
+            }
+#pragma unroll
+            for (int i = 0; i < AI; ++i)
+#pragma unroll
+                for (int j = 0; j < BJ; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < AI; ++i)
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr + i * 16 + (lane >> 4) * 4 + r;
+                int col = n0 + wc + j * 16 + (lane & 15);
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_bf16_256x256_d2_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
+    gemm_big2_body<256, 256, 4, 2, 32>(A, Bt, C, M, N, K);
+}

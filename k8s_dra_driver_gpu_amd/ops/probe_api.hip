@@ -51,6 +51,11 @@ extern "C" __global__ void gemm_bf16_128_pipe_kernel(const short*, const short*,
 extern "C" __global__ void gemm_bf16_128_pipe_bk64_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_128_pipe2_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_128_pipe3_kernel(const short*, const short*, float*, int, int, int);
+extern "C" __global__ void gemm_bf16_128_pipe2r_kernel(const short*, const short*, float*, int, int, int);
+extern "C" __global__ void gemm_bf16_256_kernel(const short*, const short*, float*, int, int, int);
+extern "C" __global__ void gemm_bf16_256x256_kernel(const short*, const short*, float*, int, int, int);
+extern "C" __global__ void gemm_bf16_256x256_d2_kernel(const short*, const short*, float*, int, int, int);
+extern "C" __global__ void gemm_bf16_256_bk64_kernel(const short*, const short*, float*, int, int, int);
 
 // bk selector shared by the gemm entry points: 32/64 pick the 16x16x32
 // tiling at that K-depth; 232/264 the 32x32x16 tiling (measured slower,
@@ -65,8 +70,20 @@ static inline void (*gemm_kern_for(int bk))(const short*, const short*, float*, 
         case 364: return gemm_bf16_128_pipe_bk64_kernel;
         case 432: return gemm_bf16_128_pipe2_kernel;
         case 532: return gemm_bf16_128_pipe3_kernel;
+        case 632: return gemm_bf16_128_pipe2r_kernel;
+        case 732: return gemm_bf16_256_kernel;
+        case 764: return gemm_bf16_256_bk64_kernel;
+        case 832: return gemm_bf16_256x256_kernel;
+        case 842: return gemm_bf16_256x256_d2_kernel;
         default:  return gemm_bf16_128_kernel;
     }
+}
+
+// launch geometry per kernel variant: (tile_m, tile_n, threads)
+static inline void gemm_geom_for(int bk, int* tm, int* tn, int* threads) {
+    if (bk == 732 || bk == 764) { *tm = 256; *tn = 128; *threads = 512; return; }
+    if (bk == 832 || bk == 842) { *tm = 256; *tn = 256; *threads = 512; return; }
+    *tm = 128; *tn = 128; *threads = 256;
 }
 extern "C" __global__ void p2p_reduce_kernel(float4v*, const float4v*, long);
 
@@ -341,17 +358,19 @@ double fp_gemm_bf16_tflops_ex(int dev, int size, int iters, int bk) {
     CHK(hipMalloc(&C, (size_t)M * N * sizeof(float)));
     CHK(hipMemset(A, 0x3c, (size_t)M * K * sizeof(short)));
     CHK(hipMemset(Bt, 0x3b, (size_t)N * K * sizeof(short)));
-    dim3 grid((M / 128) * (N / 128));
+    int tm, tn, threads;
+    gemm_geom_for(bk, &tm, &tn, &threads);
+    dim3 grid((M / tm) * (N / tn));
     auto kern = gemm_kern_for(bk);
     hipEvent_t t0, t1;
     CHK(hipEventCreate(&t0));
     CHK(hipEventCreate(&t1));
-    hipLaunchKernelGGL(kern, grid, dim3(256), 0, 0, A, Bt, C, M, N, K);
+    hipLaunchKernelGGL(kern, grid, dim3(threads), 0, 0, A, Bt, C, M, N, K);
     CHK(hipGetLastError());
     CHK(hipDeviceSynchronize());
     CHK(hipEventRecord(t0));
     for (int i = 0; i < iters; ++i)
-        hipLaunchKernelGGL(kern, grid, dim3(256), 0, 0, A, Bt, C, M, N, K);
+        hipLaunchKernelGGL(kern, grid, dim3(threads), 0, 0, A, Bt, C, M, N, K);
     CHK(hipEventRecord(t1));
     CHK(hipEventSynchronize(t1));
     double ms = time_kernel_ms(t0, t1);
@@ -364,14 +383,20 @@ double fp_gemm_bf16_tflops_ex(int dev, int size, int iters, int bk) {
 }
 
 double fp_gemm_bf16_tflops(int dev, int size, int iters) {
-    // measured pipeline-depth ladder (profiles/, same-box pairs):
-    //   2-buf vmcnt(0)     851 @4096^3 / 897 @8192^3 (best of BK=32/64)
-    //   3-buf depth-1      859-862     / 901-916
-    //   4-buf depth-2      896-903     / 899-918   <- default (bk=432)
-    //   5-buf depth-3      895         / 846  (prefetch too far: L2 pollution)
-    // Depth-2 keeps 2 workgroups/CU (64 KiB LDS) and two tiles in flight
-    // across each counted-vmcnt raw barrier.
-    return fp_gemm_bf16_tflops_ex(dev, size, iters, 432);
+    // measured ladder (profiles/, same-box pairs), TF @4096^3 / @8192^3:
+    //   round 1 (128x128 tile, 4 waves):
+    //     2-buf vmcnt(0)     851 / 897   3-buf depth-1  859-862 / 901-916
+    //     4-buf depth-2      896-903 / 899-918    5-buf depth-3  895 / 846
+    //   round 2 (big tiles, 8 waves, register-hoisted fragments):
+    //     256x128 depth-1 (732)  1028-1055 / 1100-1187
+    //     256x128 BK=64   (764)   691 / 742  (VGPR/unroll collapse, as r1)
+    //     256x256 depth-1 (832)  1045-1080 / 1137-1187
+    //     256x256 depth-2 (842)  1058 / 1174  <- default
+    // The 256x256 tile halves LDS reads per MFMA (each wave 64x128 = 4x8
+    // fragments, af[4]/bf[8] hoisted) and runs 1 WG/CU (4 x 32 KiB LDS,
+    // 128 KiB of the 160 KiB budget) with two tiles in flight across each
+    // counted-vmcnt raw barrier.
+    return fp_gemm_bf16_tflops_ex(dev, size, iters, 842);
 }
 
 int fp_gemm_bf16_host_ex(int dev, const unsigned short* A, const unsigned short* Bt,
@@ -384,9 +409,11 @@ int fp_gemm_bf16_host_ex(int dev, const unsigned short* A, const unsigned short*
     CHKI(hipMalloc(&dC, (size_t)M * N * sizeof(float)));
     CHKI(hipMemcpy(dA, A, (size_t)M * K * sizeof(short), hipMemcpyHostToDevice));
     CHKI(hipMemcpy(dB, Bt, (size_t)N * K * sizeof(short), hipMemcpyHostToDevice));
-    dim3 grid(((M + 127) / 128) * ((N + 127) / 128));
+    int tm, tn, threads;
+    gemm_geom_for(bk, &tm, &tn, &threads);
+    dim3 grid(((M + tm - 1) / tm) * ((N + tn - 1) / tn));
     auto kern = gemm_kern_for(bk);
-    hipLaunchKernelGGL(kern, grid, dim3(256), 0, 0, dA, dB, dC, M, N, K);
+    hipLaunchKernelGGL(kern, grid, dim3(threads), 0, 0, dA, dB, dC, M, N, K);
     CHKI(hipGetLastError());
     CHKI(hipDeviceSynchronize());
     CHKI(hipMemcpy(C, dC, (size_t)M * N * sizeof(float), hipMemcpyDeviceToHost));

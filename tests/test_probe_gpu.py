@@ -249,6 +249,27 @@ class TestGemmProbe:
             err = np.abs(d - ref).max() / max(1.0, np.abs(ref).max())
             assert err < 2e-3, f"bk={bk}: rel err {err}"
 
+    def test_gemm_bigtile_numerics_vs_torch_fp32(self, probe):
+        """Round-2 big-tile kernels (256x128 depth-1 = 732; 256x256 depth-1/2
+        = 832/842, the default dispatch): register-hoisted fragments + 8-wave
+        workgroups — verified with many K-steps and non-square shapes."""
+        import torch
+
+        rng = np.random.default_rng(17)
+        for M, N, K, bk in ((512, 128, 160, 732), (256, 384, 224, 732),
+                            (512, 512, 192, 832), (512, 256, 224, 842),
+                            (1024, 1024, 1024, 732), (1024, 1024, 1024, 832),
+                            (1024, 1024, 1024, 842)):
+            a = rng.standard_normal((M, K), dtype=np.float32)
+            bt = rng.standard_normal((N, K), dtype=np.float32)
+            d = probe.gemm_bf16(a, bt, bk=bk)
+            ref = (
+                torch.from_numpy(probe.bf16_truncate(a))
+                @ torch.from_numpy(probe.bf16_truncate(bt)).T
+            ).numpy()
+            err = np.abs(d - ref).max() / max(1.0, np.abs(ref).max())
+            assert err < 2e-3, f"bk={bk}: rel err {err}"
+
     def test_gemm_mfma32_numerics_vs_torch_fp32(self, probe):
         """Same contract for the 32x32x16-tiling variant (bk=232/264)."""
         import torch
