@@ -1,0 +1,30 @@
+"""Grouped-swizzle sweep: fp8 (v 2/24/28/216) and bf16 (842/844/848)."""
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import numpy as np
+from k8s_dra_driver_gpu_amd.fabric import probe
+
+rng = np.random.default_rng(5)
+M, N, K = 512, 384, 256
+a = rng.standard_normal((M, K)).astype(np.float32)
+bt = rng.standard_normal((N, K)).astype(np.float32)
+ref8 = (probe.fp8_e4m3_to_f32(probe.to_fp8_e4m3(a)).astype(np.float64)
+        @ probe.fp8_e4m3_to_f32(probe.to_fp8_e4m3(bt)).astype(np.float64).T)
+for v in (24, 28, 216):
+    d = probe.gemm_fp8(a, bt, variant=v)
+    err = np.abs(d - ref8).max() / np.abs(ref8).max()
+    print(f"fp8 v={v} rel_err={err:.2e}", "OK" if err < 1e-2 else "FAIL")
+refb = probe.bf16_truncate(a) @ probe.bf16_truncate(bt).T
+for bk in (844, 848):
+    d = probe.gemm_bf16(a, bt, bk=bk)
+    err = np.abs(d - refb).max() / np.abs(refb).max()
+    print(f"bf16 bk={bk} rel_err={err:.2e}", "OK" if err < 1e-2 else "FAIL")
+sys.stdout.flush()
+
+for size, iters in ((4096, 10), (8192, 5)):
+    for v in (2, 24, 28, 216):
+        tf = probe.gemm_fp8_tflops_ex(0, size, iters, v)
+        print(f"fp8 size={size} v={v}: {tf:.0f} TF"); sys.stdout.flush()
+    for bk in (842, 844, 848):
+        tf = probe.gemm_bf16_tflops_ex(0, size, iters, bk)
+        print(f"bf16 size={size} bk={bk}: {tf:.0f} TF"); sys.stdout.flush()

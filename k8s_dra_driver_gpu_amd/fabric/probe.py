@@ -42,6 +42,28 @@ def _load() -> ctypes.CDLL:
     lib.fp_hbm_copy_gbps.argtypes = [ctypes.c_int, ctypes.c_size_t, ctypes.c_int]
     lib.fp_mfma_bf16_tflops.restype = ctypes.c_double
     lib.fp_mfma_bf16_tflops.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int]
+    lib.fp_mfma_fp8_tflops.restype = ctypes.c_double
+    lib.fp_mfma_fp8_tflops.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int]
+    lib.fp_gemm_fp8_tflops_ex.restype = ctypes.c_double
+    lib.fp_gemm_fp8_tflops_ex.argtypes = [
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int]
+    lib.fp_gemm_fp8_host_ex.restype = ctypes.c_int
+    lib.fp_gemm_fp8_host_ex.argtypes = [
+        ctypes.c_int,
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_float),
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+    ]
+    lib.fp_mfma_fp8_tile_gemm_host.restype = ctypes.c_int
+    lib.fp_mfma_fp8_tile_gemm_host.argtypes = [
+        ctypes.c_int,
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_float),
+        ctypes.c_int,
+        ctypes.c_int,
+    ]
     lib.fp_mfma_tile_gemm_host.restype = ctypes.c_int
     lib.fp_mfma_tile_gemm_host.argtypes = [
         ctypes.c_int,
@@ -195,8 +217,8 @@ def gemm_bf16(a: np.ndarray, bt: np.ndarray, dev: int = 0, bk: int = 32) -> np.n
     # selector -> K-step depth: 232/264 = 32x32x16 tiling, 332/364 = the
     # 3-buffer pipelined kernel (counted vmcnt + raw barrier)
     kstep = {32: 32, 64: 64, 232: 32, 264: 64, 332: 32, 364: 64, 432: 32,
-             532: 32, 632: 32, 732: 32, 764: 64, 832: 32, 842: 32}[bk]
-    tile_m = {732: 256, 764: 256, 832: 256, 842: 256}.get(bk, 128)
+             532: 32, 632: 32, 732: 32, 764: 64, 832: 32, 842: 32, 844: 32, 848: 32}[bk]
+    tile_m = {732: 256, 764: 256, 832: 256, 842: 256, 844: 256, 848: 256}.get(bk, 128)
     assert K == K2 and M % tile_m == 0 and N % 128 == 0 and K % kstep == 0
     a_bf = _to_bf16_bits(np.ascontiguousarray(a, dtype=np.float32))
     b_bf = _to_bf16_bits(np.ascontiguousarray(bt, dtype=np.float32))
@@ -222,6 +244,111 @@ def burn(dev: int = 0, duration_ms: int = 2000) -> tuple:
     if rc < 0:
         raise ProbeError(f"burn failed with hip error {-rc}")
     return tf.value, gb.value
+
+
+def to_fp8_e4m3(x: np.ndarray) -> np.ndarray:
+    """Encode float32 -> OCP fp8 e4m3fn bytes (round-to-nearest-even,
+    saturate to +-448, no inf; gfx950 uses OCP, NOT the MI300X fnuz
+    encoding — cdna_hip_programming.md "FP8 — OCP, not FNUZ")."""
+    x = np.asarray(x, dtype=np.float32)
+    out = np.zeros(x.shape, dtype=np.uint8)
+    sign = (x < 0) | ((x == 0) & (np.signbit(x)))
+    ax = np.abs(x)
+    ax = np.minimum(ax, 448.0)  # saturate (e4m3fn has no inf)
+    # normals: exponent range [-6, 8]; subnormals below 2^-6
+    e = np.floor(np.log2(np.maximum(ax, 1e-45))).astype(np.int32)
+    e = np.clip(e, -6, 8)
+    scale = np.exp2(e.astype(np.float64) - 3)  # mantissa step 2^(e-3)
+    q = np.rint(ax / scale)  # RNE via rint
+    # mantissa overflow (q == 16) bumps the exponent
+    bump = q >= 16
+    e = e + bump.astype(np.int32)
+    q = np.where(bump, 8, q)  # 16/2 -> 1.0 -> stored mantissa 0 + implied 1 -> q=8
+    over = e > 8
+    e = np.where(over, 8, e)
+    q = np.where(over, 15, q)  # clamp to max normal 448 = 1.75 * 2^8
+    is_sub = ax < 2 ** -6
+    # normal: bits = ((e+7)<<3) | (q-8); subnormal: e field 0, mantissa = round(ax/2^-9)
+    sub_m = np.clip(np.rint(ax / 2.0 ** -9), 0, 7).astype(np.uint8)
+    norm_bits = (((e + 7) << 3) | (q.astype(np.int32) - 8)).astype(np.uint8)
+    out = np.where(is_sub, sub_m, norm_bits)
+    out = np.where(ax == 0, 0, out)
+    out |= (sign.astype(np.uint8) << 7)
+    return out.astype(np.uint8)
+
+
+def fp8_e4m3_to_f32(b: np.ndarray) -> np.ndarray:
+    """Decode OCP e4m3fn bytes -> float32 (the dequant reference)."""
+    b = np.asarray(b, dtype=np.uint8)
+    sign = np.where(b & 0x80, -1.0, 1.0).astype(np.float32)
+    e = ((b >> 3) & 0xF).astype(np.int32)
+    m = (b & 0x7).astype(np.float32)
+    normal = e > 0
+    val = np.where(normal,
+                   (1.0 + m / 8.0) * np.exp2((e - 7).astype(np.float32)),
+                   (m / 8.0) * np.exp2(np.float32(-6)))
+    return (sign * val).astype(np.float32)
+
+
+def mfma_fp8_tile_gemm(a: np.ndarray, b: np.ndarray, dev: int = 0,
+                       layout: int = 0) -> np.ndarray:
+    """D[16,16] = a[16,K] @ b[K,16] on the MX-fp8 (e4m3, scale=1) matrix
+    cores via mfma_scale_f32_16x16x128_f8f6f4. `layout` selects the A/B
+    fragment-map hypothesis (see fabric_probe.hip) — pinned empirically."""
+    K = a.shape[1]
+    assert a.shape == (16, K) and b.shape == (K, 16) and K % 128 == 0
+    a8 = to_fp8_e4m3(np.ascontiguousarray(a, dtype=np.float32))
+    b8 = to_fp8_e4m3(np.ascontiguousarray(b, dtype=np.float32))
+    out = np.zeros((16, 16), dtype=np.float32)
+    rc = _load().fp_mfma_fp8_tile_gemm_host(
+        dev,
+        a8.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        b8.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+        K, layout,
+    )
+    if rc < 0:
+        raise ProbeError(f"mfma_fp8_tile_gemm failed with hip error {-rc}")
+    return out
+
+
+def mfma_fp8_tflops(dev: int = 0, inner_iters: int = 2048, launches: int = 20) -> float:
+    """Register-resident mfma_scale fp8 issue-rate ceiling (no memory)."""
+    return _check(_load().fp_mfma_fp8_tflops(dev, inner_iters, launches), "mfma_fp8")
+
+
+def gemm_fp8_tflops_ex(dev: int = 0, size: int = 4096, iters: int = 10,
+                       variant: int = 316) -> float:
+    """MX-fp8 (e4m3, scale=1) GEMM throughput. Variants: 1 = 128x128;
+    2/24/28/216/232 = 256x128 3-buf (suffix = tile-group swizzle);
+    3/316 = 256x256 2-buf dual-barrier (champion: 316 = G16 swizzle,
+    measured 1469/1625 TF @4096^3/8192^3 vs the 4780 TF mfma_scale
+    ceiling — gpurun_out/r2s15)."""
+    return _check(_load().fp_gemm_fp8_tflops_ex(dev, size, iters, variant),
+                  "gemm_fp8_ex")
+
+
+def gemm_fp8(a: np.ndarray, bt: np.ndarray, dev: int = 0,
+             variant: int = 316) -> np.ndarray:
+    """C[M,N] = a[M,K] @ bt[N,K]^T on the MX-fp8 GEMM kernel (e4m3 in,
+    fp32 out); inputs are float32, quantized to e4m3 exactly as consumed."""
+    M, K = a.shape
+    N, K2 = bt.shape
+    tile_m = 128 if variant == 1 else 256  # all 2xx/3xx variants use 256
+    assert K == K2 and M % tile_m == 0 and N % 128 == 0 and K % 128 == 0
+    a8 = to_fp8_e4m3(np.ascontiguousarray(a, dtype=np.float32))
+    b8 = to_fp8_e4m3(np.ascontiguousarray(bt, dtype=np.float32))
+    out = np.zeros((M, N), dtype=np.float32)
+    rc = _load().fp_gemm_fp8_host_ex(
+        dev,
+        a8.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        b8.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+        M, N, K, variant,
+    )
+    if rc < 0:
+        raise ProbeError(f"gemm_fp8 failed with hip error {-rc}")
+    return out
 
 
 def _to_bf16_bits(x: np.ndarray) -> np.ndarray:

@@ -255,6 +255,88 @@ mfma_bf16_tile_gemm_kernel(const short* __restrict__ A,
 }
 
 // ---------------------------------------------------------------------------
+// MX-fp8 (OCP e4m3fn) 16x16x128 block-scaled MFMA tile verify. gfx950's
+// mfma_scale_f32_16x16x128_f8f6f4 is the only large-K fp8 MFMA (2x the
+// bf16 rate; cdna_hip_programming.md gfx950 intrinsic list). Scales are
+// E8M0, one per 32-element MX block = one byte per lane per operand; 0x7F
+// (2^0) makes the scaled form numerically plain fp8.
+//
+// `layout` selects the A/B lane->element hypothesis so the real mapping is
+// pinned EMPIRICALLY in one GPU call (no ISA doc for the fragment map in
+// this image):
+//   0: k = (lane>>4)*32 + e            (natural extension of bf16 16x16x32)
+//   1: k = e*4 + (lane>>4)             (K-interleaved across lane groups)
+//   2: k = (lane>>4)*8 + (e&7) + (e>>3)*32   (8-elem dwords from 4 blocks)
+// C/D layout is shape-determined and dtype-independent on gfx950 (guide
+// "Fragment layout"): row=(lane>>4)*4+r, col=lane&15, same as bf16.
+// ---------------------------------------------------------------------------
+
+typedef int i32x8 __attribute__((ext_vector_type(8)));
+
+extern "C" __global__ void __launch_bounds__(WAVE)
+mfma_fp8_tile_gemm_kernel(const unsigned char* __restrict__ A,
+                          const unsigned char* __restrict__ B,
+                          float* __restrict__ D, int K, int layout) {
+    int lane = threadIdx.x & (WAVE - 1);
+    int row = lane & 15;
+    int kgrp = lane >> 4;  // 0..3
+    f32x4 acc = {};
+    for (int k0 = 0; k0 < K; k0 += 128) {
+        union {
+            i32x8 v;
+            unsigned char b[32];
+        } a, bb;
+        for (int e = 0; e < 32; ++e) {
+            int k;
+            if (layout == 1)
+                k = e * 4 + kgrp;
+            else if (layout == 2)
+                k = kgrp * 8 + (e & 7) + (e >> 3) * 32;
+            else
+                k = kgrp * 32 + e;
+            a.b[e] = A[row * K + k0 + k];   // A is [16][K] row-major
+            bb.b[e] = B[(k0 + k) * 16 + row];  // B is [K][16]; col = row idx
+        }
+        acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            a.v, bb.v, acc, 0 /*cbsz=fp8*/, 0 /*blgp=fp8*/,
+            0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int out_row = kgrp * 4 + r;
+        D[out_row * 16 + row] = acc[r];
+    }
+}
+
+// Register-resident mfma_scale_f32_16x16x128_f8f6f4 loop: the fp8 MFMA
+// issue-rate ceiling (no memory traffic). 4 independent accumulators per
+// lane hide the MFMA latency, same shape as mfma_bf16_loop_kernel.
+extern "C" __global__ void __launch_bounds__(PROBE_BLOCK)
+mfma_fp8_loop_kernel(const int* __restrict__ seed, float* __restrict__ sink,
+                     int iters) {
+    int lane = threadIdx.x & (WAVE - 1);
+    union { i32x8 v; int i[8]; } a, b;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        a.i[e] = seed[(lane * 8 + e) & 1023];
+        b.i[e] = seed[(lane * 8 + e + 512) & 1023];
+    }
+    f32x4 acc0 = {}, acc1 = {}, acc2 = {}, acc3 = {};
+    for (int it = 0; it < iters; ++it) {
+        acc0 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            a.v, b.v, acc0, 0, 0, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+        acc1 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            a.v, b.v, acc1, 0, 0, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+        acc2 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            a.v, b.v, acc2, 0, 0, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+        acc3 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            a.v, b.v, acc3, 0, 0, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+    }
+    float r = acc0[0] + acc1[1] + acc2[2] + acc3[3];
+    if (r == -1.0f) sink[0] = r;
+}
+
+// ---------------------------------------------------------------------------
 // xGMI p2p pull probe: read from a peer GPU's buffer (mapped via
 // hipDeviceEnablePeerAccess) into local HBM. Bandwidth is bound by the xGMI
 // links to that peer (~153 GB/s per link).

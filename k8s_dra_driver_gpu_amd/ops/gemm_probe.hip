@@ -982,7 +982,7 @@ gemm_bf16_256_bk64_kernel(const short* A, const short* Bt, float* C, int M, int 
 // Depth-2 variant of the generic big-tile body: 4 LDS buffers, TWO tiles in
 // flight across each barrier (pipe2-style counted vmcnt). For 256x256 this
 // is 4 x 32 KiB = 128 KiB LDS (fits the 160 KiB CU budget at 1 WG/CU).
-template <int TBM, int TBN, int TWR, int TWC, int BK>
+template <int TBM, int TBN, int TWR, int TWC, int BK, int GROUP = 1>
 __device__ __forceinline__ void gemm_big2_body(
     const short* __restrict__ A, const short* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
@@ -997,8 +997,22 @@ __device__ __forceinline__ void gemm_big2_body(
     auto ldsB = [&](int buf) -> short* { return lds + buf * HALF + TBM * BK; };
 
     const int tiles_n = (N + TBN - 1) / TBN;
-    const int m0 = (blockIdx.x / tiles_n) * TBM;
-    const int n0 = (blockIdx.x % tiles_n) * TBN;
+    int tile_m, tile_n;
+    if (GROUP > 1) {
+        // grouped swizzle for L2 reuse (see gemm_fp8_body)
+        const int tiles_m = (M + TBM - 1) / TBM;
+        const int per_group = GROUP * tiles_n;
+        const int gid = blockIdx.x / per_group;
+        const int first_m = gid * GROUP;
+        const int gsz = min(GROUP, tiles_m - first_m);
+        tile_m = first_m + (blockIdx.x % per_group) % gsz;
+        tile_n = (blockIdx.x % per_group) / gsz;
+    } else {
+        tile_m = blockIdx.x / tiles_n;
+        tile_n = blockIdx.x % tiles_n;
+    }
+    const int m0 = tile_m * TBM;
+    const int n0 = tile_n * TBN;
 
     const int tid = threadIdx.x;
     const int lane = tid & (WAVE - 1);
@@ -1097,4 +1111,317 @@ __device__ __forceinline__ void gemm_big2_body(
 extern "C" __global__ void __launch_bounds__(512)
 gemm_bf16_256x256_d2_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
     gemm_big2_body<256, 256, 4, 2, 32>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_bf16_256x256_d2_g4_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
+    gemm_big2_body<256, 256, 4, 2, 32, 4>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_bf16_256x256_d2_g8_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
+    gemm_big2_body<256, 256, 4, 2, 32, 8>(A, Bt, C, M, N, K);
+}
+
+// ---------------------------------------------------------------------------
+// MX-fp8 (OCP e4m3, scale=1) GEMM — mfma_scale_f32_16x16x128_f8f6f4 runs at
+// 2x the bf16 MFMA rate (the MI355X ~5 PF headline path). Same structure as
+// the bf16 big-tile bodies: glds staging, counted-vmcnt 3-buffer pipeline,
+// register-hoisted fragments. BK = 128 (one MFMA K-depth per step); LDS
+// tile rows are 128 B = 8 x 16 B chunks, elements are 1 B so staged bytes
+// per FLOP are HALF the bf16 kernel's.
+//
+// Fragment k-mapping note (measured, gpurun_out/r2s12/fp8_layout.txt): the
+// matmul is invariant to any k-permutation applied consistently to both
+// A and B fragments, so the natural extension map (lane kgrp*32 + e,
+// 32 contiguous bytes per lane) is used for both operands; with
+// per-block scales all = 0x7F (2^0) the MX blocking cannot introduce a
+// scale/k mismatch.
+// ---------------------------------------------------------------------------
+
+typedef int i32x8 __attribute__((ext_vector_type(8)));
+
+template <int TBM, int TBN, int TWR, int TWC, int GROUP = 1>
+__device__ __forceinline__ void gemm_fp8_body(
+    const unsigned char* __restrict__ A,   // [M][K] row-major e4m3
+    const unsigned char* __restrict__ Bt,  // [N][K] row-major e4m3
+    float* __restrict__ C,                 // [M][N] row-major f32
+    int M, int N, int K) {
+    constexpr int BK = 128;
+    constexpr int NT = TWR * TWC * WAVE;
+    constexpr int WM = TBM / TWR;
+    constexpr int WN = TBN / TWC;
+    constexpr int AI = WM / 16;
+    constexpr int BJ = WN / 16;
+    __shared__ unsigned char lds[3 * (TBM * BK + TBN * BK)];
+    const int HALF = TBM * BK + TBN * BK;
+    auto ldsA = [&](int buf) -> unsigned char* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> unsigned char* { return lds + buf * HALF + TBM * BK; };
+
+    const int tiles_n = (N + TBN - 1) / TBN;
+    int tile_m, tile_n;
+    if (GROUP > 1) {
+        // grouped swizzle (L2 reuse): concurrent blocks cover only GROUP
+        // m-tiles, so each B column-panel is reused GROUP times and the A
+        // working set stays inside the per-XCD L2 while both kernels are
+        // HBM-bound at large sizes
+        const int tiles_m = (M + TBM - 1) / TBM;
+        const int per_group = GROUP * tiles_n;
+        const int gid = blockIdx.x / per_group;
+        const int first_m = gid * GROUP;
+        const int gsz = min(GROUP, tiles_m - first_m);
+        tile_m = first_m + (blockIdx.x % per_group) % gsz;
+        tile_n = (blockIdx.x % per_group) / gsz;
+    } else {
+        tile_m = blockIdx.x / tiles_n;
+        tile_n = blockIdx.x % tiles_n;
+    }
+    const int m0 = tile_m * TBM;
+    const int n0 = tile_n * TBN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid / TWC) * WM;
+    const int wc = (wid % TWC) * WN;
+
+    constexpr int ACH = TBM * (BK / 16);  // 16 B chunks per A tile
+    constexpr int APH = ACH / NT;
+    constexpr int BCH = TBN * (BK / 16);
+    constexpr int BPH = BCH / NT;
+    static_assert(APH * NT == ACH && BPH * NT == BCH, "phase split");
+    constexpr int S = APH + BPH;
+    auto stage = [&](int buf, int k0) {
+        unsigned char* la = ldsA(buf);
+        unsigned char* lb = ldsB(buf);
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 16);
+            int c = chunk % (BK / 16);
+            const unsigned char* ga = &A[(size_t)(m0 + r) * K + k0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+#pragma unroll
+        for (int phase = 0; phase < BPH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 16);
+            int c = chunk % (BK / 16);
+            const unsigned char* gb = &Bt[(size_t)(n0 + r) * K + k0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+    };
+
+    f32x4 acc[AI][BJ] = {};
+    const int kg = (lane >> 4) * 32;  // this lane's 32-byte k-slice
+    const int steps = K / BK;
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s % 3;
+        if (s + 1 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 2 < steps) stage((s + 2) % 3, (s + 2) * BK);
+
+        const unsigned char* la = ldsA(buf);
+        const unsigned char* lb = ldsB(buf);
+        i32x8 af[AI], bf[BJ];
+#pragma unroll
+        for (int i = 0; i < AI; ++i) {
+            const int ar = wr + i * 16 + (lane & 15);
+            af[i] = *(const i32x8*)&la[ar * BK + kg];
+        }
+#pragma unroll
+        for (int j = 0; j < BJ; ++j) {
+            const int bc = wc + j * 16 + (lane & 15);
+            bf[j] = *(const i32x8*)&lb[bc * BK + kg];
+        }
+#pragma unroll
+        for (int i = 0; i < AI; ++i)
+#pragma unroll
+            for (int j = 0; j < BJ; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                    af[i], bf[j], acc[i][j], 0, 0, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+    }
+
+#pragma unroll
+    for (int i = 0; i < AI; ++i)
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr + i * 16 + (lane >> 4) * 4 + r;
+                int col = n0 + wc + j * 16 + (lane & 15);
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+// 128x128: 3 x 32 KiB LDS (2 WG/CU), 4 waves of 64x64
+extern "C" __global__ void __launch_bounds__(256)
+gemm_fp8_128_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                    int M, int N, int K) {
+    gemm_fp8_body<128, 128, 2, 2>(A, Bt, C, M, N, K);
+}
+
+// 256x128: 3 x 48 KiB LDS (1 WG/CU), 8 waves of 64x64
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_256_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                    int M, int N, int K) {
+    gemm_fp8_body<256, 128, 4, 2>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_256_g4_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                       int M, int N, int K) {
+    gemm_fp8_body<256, 128, 4, 2, 4>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_256_g8_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                       int M, int N, int K) {
+    gemm_fp8_body<256, 128, 4, 2, 8>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_256_g16_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                        int M, int N, int K) {
+    gemm_fp8_body<256, 128, 4, 2, 16>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_256_g32_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                        int M, int N, int K) {
+    gemm_fp8_body<256, 128, 4, 2, 32>(A, Bt, C, M, N, K);
+}
+
+// 256x256 fp8, 2 x 64 KiB LDS double buffer (dual barrier per step since a
+// staged buffer is immediately reused), 8 waves of 64x128 (AI=4, BJ=8)
+template <int GROUP>
+__device__ __forceinline__ void gemm_fp8_256x256_body(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+    constexpr int TBM = 256, TBN = 256, BK = 128, NT = 512;
+    constexpr int AI = 4, BJ = 8;
+    __shared__ unsigned char lds[2 * (TBM * BK + TBN * BK)];
+    const int HALF = TBM * BK + TBN * BK;
+    auto ldsA = [&](int buf) -> unsigned char* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> unsigned char* { return lds + buf * HALF + TBM * BK; };
+
+    const int tiles_n = (N + TBN - 1) / TBN;
+    int tile_m, tile_n;
+    if (GROUP > 1) {
+        const int tiles_m = (M + TBM - 1) / TBM;
+        const int per_group = GROUP * tiles_n;
+        const int gid = blockIdx.x / per_group;
+        const int first_m = gid * GROUP;
+        const int gsz = min(GROUP, tiles_m - first_m);
+        tile_m = first_m + (blockIdx.x % per_group) % gsz;
+        tile_n = (blockIdx.x % per_group) / gsz;
+    } else {
+        tile_m = blockIdx.x / tiles_n;
+        tile_n = blockIdx.x % tiles_n;
+    }
+    const int m0 = tile_m * TBM;
+    const int n0 = tile_n * TBN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid >> 1) * 64;
+    const int wc = (wid & 1) * 128;
+
+    constexpr int ACH = TBM * (BK / 16);
+    constexpr int APH = ACH / NT;   // 4
+    auto stage = [&](int buf, int k0) {
+        unsigned char* la = ldsA(buf);
+        unsigned char* lb = ldsB(buf);
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 16);
+            int c = chunk % (BK / 16);
+            const unsigned char* ga = &A[(size_t)(m0 + r) * K + k0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+            const unsigned char* gb = &Bt[(size_t)(n0 + r) * K + k0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+    };
+
+    f32x4 acc[AI][BJ] = {};
+    const int kg = (lane >> 4) * 32;
+    const int steps = K / BK;
+
+    stage(0, 0);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s & 1;
+        // stage(s) fully landed
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        // prefetch s+1 into the other buffer UNDER this step's MFMAs —
+        // safe: compute(s-1) finished before the barrier above
+        if (s + 1 < steps) stage(buf ^ 1, (s + 1) * BK);
+
+        const unsigned char* la = ldsA(buf);
+        const unsigned char* lb = ldsB(buf);
+        i32x8 af[AI], bf[BJ];
+#pragma unroll
+        for (int i = 0; i < AI; ++i) {
+            const int ar = wr + i * 16 + (lane & 15);
+            af[i] = *(const i32x8*)&la[ar * BK + kg];
+        }
+#pragma unroll
+        for (int j = 0; j < BJ; ++j) {
+            const int bc = wc + j * 16 + (lane & 15);
+            bf[j] = *(const i32x8*)&lb[bc * BK + kg];
+        }
+#pragma unroll
+        for (int i = 0; i < AI; ++i)
+#pragma unroll
+            for (int j = 0; j < BJ; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                    af[i], bf[j], acc[i][j], 0, 0, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+        // every wave done reading buf before the next stage overwrites it
+        __syncthreads();
+    }
+
+#pragma unroll
+    for (int i = 0; i < AI; ++i)
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr + i * 16 + (lane >> 4) * 4 + r;
+                int col = n0 + wc + j * 16 + (lane & 15);
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_256x256_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                        int M, int N, int K) {
+    gemm_fp8_256x256_body<1>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_256x256_g16_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                            int M, int N, int K) {
+    gemm_fp8_256x256_body<16>(A, Bt, C, M, N, K);
 }

@@ -40,7 +40,9 @@ extern "C" __global__ void hbm_copy_nt_kernel(float4v*, const float4v*, long);
 extern "C" __global__ void hbm_copy_kernel(float4v*, const float4v*, long);
 extern "C" __global__ void hbm_block_sum_kernel(const float*, float*, long);
 extern "C" __global__ void mfma_bf16_loop_kernel(const short*, float*, int);
+extern "C" __global__ void mfma_fp8_loop_kernel(const int*, float*, int);
 extern "C" __global__ void mfma_bf16_tile_gemm_kernel(const short*, const short*, float*, int);
+extern "C" __global__ void mfma_fp8_tile_gemm_kernel(const unsigned char*, const unsigned char*, float*, int, int);
 extern "C" __global__ void p2p_read_kernel(float4v*, const float4v*, long);
 extern "C" __global__ void vmfault_kernel(float*);
 extern "C" __global__ void gemm_bf16_128_kernel(const short*, const short*, float*, int, int, int);
@@ -56,6 +58,16 @@ extern "C" __global__ void gemm_bf16_256_kernel(const short*, const short*, floa
 extern "C" __global__ void gemm_bf16_256x256_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256x256_d2_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256_bk64_kernel(const short*, const short*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_128_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_256_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_256_g4_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_256_g8_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_256_g16_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_256_g32_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_256x256_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_256x256_g16_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_bf16_256x256_d2_g4_kernel(const short*, const short*, float*, int, int, int);
+extern "C" __global__ void gemm_bf16_256x256_d2_g8_kernel(const short*, const short*, float*, int, int, int);
 
 // bk selector shared by the gemm entry points: 32/64 pick the 16x16x32
 // tiling at that K-depth; 232/264 the 32x32x16 tiling (measured slower,
@@ -75,6 +87,8 @@ static inline void (*gemm_kern_for(int bk))(const short*, const short*, float*, 
         case 764: return gemm_bf16_256_bk64_kernel;
         case 832: return gemm_bf16_256x256_kernel;
         case 842: return gemm_bf16_256x256_d2_kernel;
+        case 844: return gemm_bf16_256x256_d2_g4_kernel;
+        case 848: return gemm_bf16_256x256_d2_g8_kernel;
         default:  return gemm_bf16_128_kernel;
     }
 }
@@ -82,7 +96,7 @@ static inline void (*gemm_kern_for(int bk))(const short*, const short*, float*, 
 // launch geometry per kernel variant: (tile_m, tile_n, threads)
 static inline void gemm_geom_for(int bk, int* tm, int* tn, int* threads) {
     if (bk == 732 || bk == 764) { *tm = 256; *tn = 128; *threads = 512; return; }
-    if (bk == 832 || bk == 842) { *tm = 256; *tn = 256; *threads = 512; return; }
+    if (bk == 832 || bk == 842 || bk == 844 || bk == 848) { *tm = 256; *tn = 256; *threads = 512; return; }
     *tm = 128; *tn = 128; *threads = 256;
 }
 extern "C" __global__ void p2p_reduce_kernel(float4v*, const float4v*, long);
@@ -533,6 +547,133 @@ int fp_mfma_tile_gemm_host(int dev, const unsigned short* A, const unsigned shor
     hipFree(dA);
     hipFree(dB);
     hipFree(dD);
+    return 0;
+}
+
+int fp_mfma_fp8_tile_gemm_host(int dev, const unsigned char* A, const unsigned char* B,
+                               float* D, int K, int layout) {
+    CHKI(hipSetDevice(dev));
+    unsigned char *dA, *dB;
+    float* dD;
+    CHKI(hipMalloc(&dA, (size_t)16 * K));
+    CHKI(hipMalloc(&dB, (size_t)K * 16));
+    CHKI(hipMalloc(&dD, 16 * 16 * sizeof(float)));
+    CHKI(hipMemcpy(dA, A, (size_t)16 * K, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dB, B, (size_t)K * 16, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(mfma_fp8_tile_gemm_kernel, dim3(1), dim3(64), 0, 0, dA, dB, dD, K, layout);
+    CHKI(hipGetLastError());
+    CHKI(hipDeviceSynchronize());
+    CHKI(hipMemcpy(D, dD, 16 * 16 * sizeof(float), hipMemcpyDeviceToHost));
+    hipFree(dA);
+    hipFree(dB);
+    hipFree(dD);
+    return 0;
+}
+
+// MX-fp8 GEMM (scale=1): variant 1 = 128x128/256t, 2 = 256x128/512t
+static inline void (*fp8_kern_for(int v))(const unsigned char*, const unsigned char*, float*, int, int, int) {
+    switch (v) {
+        case 2:  return gemm_fp8_256_kernel;
+        case 24: return gemm_fp8_256_g4_kernel;
+        case 28: return gemm_fp8_256_g8_kernel;
+        case 216: return gemm_fp8_256_g16_kernel;
+        case 232: return gemm_fp8_256_g32_kernel;
+        case 3:   return gemm_fp8_256x256_kernel;
+        case 316: return gemm_fp8_256x256_g16_kernel;
+        default: return gemm_fp8_128_kernel;
+    }
+}
+static inline void fp8_geom_for(int v, int* tm, int* tn, int* threads) {
+    if (v == 2 || v == 24 || v == 28 || v == 216 || v == 232) {
+        *tm = 256; *tn = 128; *threads = 512; return;
+    }
+    if (v == 3 || v == 316) { *tm = 256; *tn = 256; *threads = 512; return; }
+    *tm = 128; *tn = 128; *threads = 256;
+}
+
+double fp_gemm_fp8_tflops_ex(int dev, int size, int iters, int variant) {
+    CHK(hipSetDevice(dev));
+    int M = size, N = size, K = size;
+    unsigned char *A, *Bt;
+    float* C;
+    CHK(hipMalloc(&A, (size_t)M * K));
+    CHK(hipMalloc(&Bt, (size_t)N * K));
+    CHK(hipMalloc(&C, (size_t)M * N * sizeof(float)));
+    CHK(hipMemset(A, 0x38, (size_t)M * K));   // e4m3 1.0
+    CHK(hipMemset(Bt, 0x30, (size_t)N * K));  // e4m3 0.5
+    int tm, tn, threads;
+    fp8_geom_for(variant, &tm, &tn, &threads);
+    dim3 grid((M / tm) * (N / tn));
+    auto kern = fp8_kern_for(variant);
+    hipEvent_t t0, t1;
+    CHK(hipEventCreate(&t0));
+    CHK(hipEventCreate(&t1));
+    hipLaunchKernelGGL(kern, grid, dim3(threads), 0, 0, A, Bt, C, M, N, K);
+    CHK(hipGetLastError());
+    CHK(hipDeviceSynchronize());
+    CHK(hipEventRecord(t0));
+    for (int i = 0; i < iters; ++i)
+        hipLaunchKernelGGL(kern, grid, dim3(threads), 0, 0, A, Bt, C, M, N, K);
+    CHK(hipEventRecord(t1));
+    CHK(hipEventSynchronize(t1));
+    double ms = time_kernel_ms(t0, t1);
+    hipFree(A);
+    hipFree(Bt);
+    hipFree(C);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    return 2.0 * M * (double)N * K * iters / (ms * 1e9);
+}
+
+double fp_mfma_fp8_tflops(int dev, int inner_iters, int launches) {
+    CHK(hipSetDevice(dev));
+    int* seed;
+    float* sink;
+    CHK(hipMalloc(&seed, 1024 * sizeof(int)));
+    CHK(hipMalloc(&sink, sizeof(float)));
+    CHK(hipMemset(seed, 0x11, 1024 * sizeof(int)));
+    hipEvent_t t0, t1;
+    CHK(hipEventCreate(&t0));
+    CHK(hipEventCreate(&t1));
+    hipLaunchKernelGGL(mfma_fp8_loop_kernel, dim3(PROBE_GRID), dim3(PROBE_BLOCK), 0, 0, seed, sink, inner_iters);
+    CHK(hipGetLastError());
+    CHK(hipDeviceSynchronize());
+    CHK(hipEventRecord(t0));
+    for (int l = 0; l < launches; ++l)
+        hipLaunchKernelGGL(mfma_fp8_loop_kernel, dim3(PROBE_GRID), dim3(PROBE_BLOCK), 0, 0, seed, sink, inner_iters);
+    CHK(hipEventRecord(t1));
+    CHK(hipEventSynchronize(t1));
+    double ms = time_kernel_ms(t0, t1);
+    double waves = (double)PROBE_GRID * PROBE_BLOCK / 64.0;
+    // 4 MFMA per inner iter, each 2*16*16*128 FLOP
+    double flops = waves * 4.0 * inner_iters * launches * 2.0 * 16 * 16 * 128;
+    hipFree(seed);
+    hipFree(sink);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    return flops / (ms * 1e9);
+}
+
+int fp_gemm_fp8_host_ex(int dev, const unsigned char* A, const unsigned char* Bt,
+                        float* C, int M, int N, int K, int variant) {
+    CHKI(hipSetDevice(dev));
+    unsigned char *dA, *dB;
+    float* dC;
+    CHKI(hipMalloc(&dA, (size_t)M * K));
+    CHKI(hipMalloc(&dB, (size_t)N * K));
+    CHKI(hipMalloc(&dC, (size_t)M * N * sizeof(float)));
+    CHKI(hipMemcpy(dA, A, (size_t)M * K, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dB, Bt, (size_t)N * K, hipMemcpyHostToDevice));
+    int tm, tn, threads;
+    fp8_geom_for(variant, &tm, &tn, &threads);
+    dim3 grid(((M + tm - 1) / tm) * ((N + tn - 1) / tn));
+    hipLaunchKernelGGL(fp8_kern_for(variant), grid, dim3(threads), 0, 0, dA, dB, dC, M, N, K);
+    CHKI(hipGetLastError());
+    CHKI(hipDeviceSynchronize());
+    CHKI(hipMemcpy(C, dC, (size_t)M * N * sizeof(float), hipMemcpyDeviceToHost));
+    hipFree(dA);
+    hipFree(dB);
+    hipFree(dC);
     return 0;
 }
 
