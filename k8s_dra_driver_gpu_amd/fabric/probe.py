@@ -44,6 +44,16 @@ def _load() -> ctypes.CDLL:
     lib.fp_mfma_bf16_tflops.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int]
     lib.fp_mfma_fp8_tflops.restype = ctypes.c_double
     lib.fp_mfma_fp8_tflops.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int]
+    lib.fp_mfma_fp4_tflops.restype = ctypes.c_double
+    lib.fp_mfma_fp4_tflops.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int]
+    lib.fp_mfma_fp4_tile_gemm_host.restype = ctypes.c_int
+    lib.fp_mfma_fp4_tile_gemm_host.argtypes = [
+        ctypes.c_int,
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_float),
+        ctypes.c_int,
+    ]
     lib.fp_gemm_fp8_tflops_ex.restype = ctypes.c_double
     lib.fp_gemm_fp8_tflops_ex.argtypes = [
         ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int]
@@ -310,6 +320,93 @@ def mfma_fp8_tile_gemm(a: np.ndarray, b: np.ndarray, dev: int = 0,
     if rc < 0:
         raise ProbeError(f"mfma_fp8_tile_gemm failed with hip error {-rc}")
     return out
+
+
+# OCP e2m1 (fp4) value table: s * {0,0.5,1,1.5,2,3,4,6}
+_FP4_VALUES = np.array([0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0], dtype=np.float32)
+
+
+def to_fp4_e2m1(x: np.ndarray) -> np.ndarray:
+    """Encode float32 -> OCP fp4 e2m1 nibbles (0..15; round-to-nearest,
+    ties toward the larger magnitude per the table midpoints)."""
+    x = np.asarray(x, dtype=np.float32)
+    ax = np.abs(x).reshape(-1)
+    idx = np.argmin(np.abs(ax[:, None] - _FP4_VALUES[None, :]), axis=1).astype(np.uint8)
+    nib = idx | (np.signbit(x.reshape(-1)).astype(np.uint8) << 3)
+    return nib.reshape(x.shape)
+
+
+def fp4_e2m1_to_f32(nib: np.ndarray) -> np.ndarray:
+    nib = np.asarray(nib, dtype=np.uint8)
+    mag = _FP4_VALUES[(nib & 0x7).reshape(-1)].reshape(nib.shape)
+    return np.where(nib & 0x8, -mag, mag).astype(np.float32)
+
+
+def _pack_nibbles(nib: np.ndarray, axis_len: int) -> np.ndarray:
+    """Pack pairs along the last axis (even = low nibble)."""
+    flat = nib.reshape(-1, axis_len)
+    return (flat[:, 0::2] | (flat[:, 1::2] << 4)).astype(np.uint8)
+
+
+def mfma_fp4_tile_gemm(a: np.ndarray, b: np.ndarray, dev: int = 0) -> np.ndarray:
+    """D[32,32] = a[32,K] @ b[K,32] on the MX-fp4 (e2m1, scale=1) matrix
+    cores via mfma_scale_f32_32x32x64_f8f6f4 (the ~10 PF headline shape)."""
+    K = a.shape[1]
+    assert a.shape == (32, K) and b.shape == (K, 32) and K % 64 == 0
+    a4 = to_fp4_e2m1(np.ascontiguousarray(a, dtype=np.float32))
+    b4 = to_fp4_e2m1(np.ascontiguousarray(b, dtype=np.float32))
+    a_packed = _pack_nibbles(a4, K)                       # [32][K/2]
+    # B packed along k: byte (k/2)*32 + col
+    b_pairs = b4.reshape(K // 2, 2, 32)
+    b_packed = (b_pairs[:, 0, :] | (b_pairs[:, 1, :] << 4)).astype(np.uint8)
+    out = np.zeros((32, 32), dtype=np.float32)
+    rc = _load().fp_mfma_fp4_tile_gemm_host(
+        dev,
+        np.ascontiguousarray(a_packed).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(b_packed).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+        K,
+    )
+    if rc < 0:
+        raise ProbeError(f"mfma_fp4_tile_gemm failed with hip error {-rc}")
+    return out
+
+
+def gemm_fp4(a: np.ndarray, bt: np.ndarray, dev: int = 0,
+             variant: int = 436) -> np.ndarray:
+    """C[M,N] = a[M,K] @ bt[N,K]^T on the MX-fp4 GEMM kernel (e2m1 packed
+    two-per-byte, fp32 out). variant 4 = 256x256 2-buf, 416 = +G16 swizzle,
+    436 = 3-buf counted pipeline G16."""
+    M, K = a.shape
+    N, K2 = bt.shape
+    assert K == K2 and M % 256 == 0 and N % 256 == 0 and K % 256 == 0
+    a4 = _pack_nibbles(to_fp4_e2m1(np.ascontiguousarray(a, dtype=np.float32)), K)
+    b4 = _pack_nibbles(to_fp4_e2m1(np.ascontiguousarray(bt, dtype=np.float32)), K)
+    out = np.zeros((M, N), dtype=np.float32)
+    rc = _load().fp_gemm_fp8_host_ex(
+        dev,
+        np.ascontiguousarray(a4).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(b4).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+        M, N, K, variant,
+    )
+    if rc < 0:
+        raise ProbeError(f"gemm_fp4 failed with hip error {-rc}")
+    return out
+
+
+def gemm_fp4_tflops_ex(dev: int = 0, size: int = 4096, iters: int = 10,
+                       variant: int = 436) -> float:
+    """MX-fp4 GEMM throughput. Champion 436 (3-buf counted G16): measured
+    2649/3197 TF @4096^3/8192^3 vs the 9074 TF mfma ceiling
+    (gpurun_out/r2s17-20)."""
+    return _check(_load().fp_gemm_fp8_tflops_ex(dev, size, iters, variant),
+                  "gemm_fp4_ex")
+
+
+def mfma_fp4_tflops(dev: int = 0, inner_iters: int = 2048, launches: int = 20) -> float:
+    """Register-resident mfma_scale fp4 32x32x64 issue-rate ceiling."""
+    return _check(_load().fp_mfma_fp4_tflops(dev, inner_iters, launches), "mfma_fp4")
 
 
 def mfma_fp8_tflops(dev: int = 0, inner_iters: int = 2048, launches: int = 20) -> float:

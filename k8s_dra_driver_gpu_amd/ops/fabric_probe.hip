@@ -336,6 +336,82 @@ mfma_fp8_loop_kernel(const int* __restrict__ seed, float* __restrict__ sink,
     if (r == -1.0f) sink[0] = r;
 }
 
+// MX-fp4 (OCP e2m1) 32x32x64 block-scaled MFMA: the ~10 PF dense headline
+// path (fp4 runs only through mfma_scale_*_f8f6f4 with FMT=4). Tile-verify
+// + register ubench. fp4 elements pack two per byte (low nibble = even
+// element); the k-map is applied consistently to A and B so GEMM numerics
+// pin correctness regardless of the device's internal k-order (see the fp8
+// layout note, gpurun_out/r2s12). C/D layout is shape-determined: for
+// 32x32 shapes row = (reg&3) + 8*(reg>>2) + 4*(lane>>5), col = lane&31.
+
+extern "C" __global__ void __launch_bounds__(WAVE)
+mfma_fp4_tile_gemm_kernel(const unsigned char* __restrict__ A,  // [32][K] e2m1 packed 2/byte, row stride K/2
+                          const unsigned char* __restrict__ B,  // [K][32] packed pairs along k: byte (k/2)*32+col
+                          float* __restrict__ D, int K) {
+    int lane = threadIdx.x & (WAVE - 1);
+    int col = lane & 31;
+    int kgrp = lane >> 5;  // 0..1, each covering 32 k's
+    f32x16 acc = {};
+    for (int k0 = 0; k0 < K; k0 += 64) {
+        union {
+            i32x8 v;
+            unsigned char b[32];
+        } a, bb;
+        for (int e = 0; e < 32; ++e) a.b[e] = 0, bb.b[e] = 0;
+        // lane covers k = k0 + kgrp*32 + j, j = 0..31; packed two per byte
+        for (int j = 0; j < 32; ++j) {
+            int k = k0 + kgrp * 32 + j;
+            // A[row=col][k]: row-major, 2 elems/byte
+            unsigned char av = A[(size_t)col * (K / 2) + k / 2];
+            unsigned char an = (k & 1) ? (av >> 4) : (av & 0xF);
+            // B[k][col]: packed along k: byte index (k/2)*32 + col
+            unsigned char bv = B[(size_t)(k / 2) * 32 + col];
+            unsigned char bn = (k & 1) ? (bv >> 4) : (bv & 0xF);
+            if (j & 1) {
+                a.b[j / 2] |= an << 4;
+                bb.b[j / 2] |= bn << 4;
+            } else {
+                a.b[j / 2] |= an;
+                bb.b[j / 2] |= bn;
+            }
+        }
+        acc = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            a.v, bb.v, acc, 4 /*cbsz=fp4*/, 4 /*blgp=fp4*/,
+            0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+    }
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+        int row = (r & 3) + 8 * (r >> 2) + 4 * kgrp;
+        D[row * 32 + col] = acc[r];
+    }
+}
+
+// Register-resident fp4 issue-rate ubench (4 independent f32x16 accumulators)
+extern "C" __global__ void __launch_bounds__(PROBE_BLOCK)
+mfma_fp4_loop_kernel(const int* __restrict__ seed, float* __restrict__ sink,
+                     int iters) {
+    int lane = threadIdx.x & (WAVE - 1);
+    union { i32x8 v; int i[8]; } a, b;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        a.i[e] = seed[(lane * 8 + e) & 1023];
+        b.i[e] = seed[(lane * 8 + e + 512) & 1023];
+    }
+    f32x16 acc0 = {}, acc1 = {}, acc2 = {}, acc3 = {};
+    for (int it = 0; it < iters; ++it) {
+        acc0 = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            a.v, b.v, acc0, 4, 4, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+        acc1 = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            a.v, b.v, acc1, 4, 4, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+        acc2 = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            a.v, b.v, acc2, 4, 4, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+        acc3 = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            a.v, b.v, acc3, 4, 4, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+    }
+    float r = acc0[0] + acc1[1] + acc2[2] + acc3[3];
+    if (r == -1.0f) sink[0] = r;
+}
+
 // ---------------------------------------------------------------------------
 // xGMI p2p pull probe: read from a peer GPU's buffer (mapped via
 // hipDeviceEnablePeerAccess) into local HBM. Bandwidth is bound by the xGMI

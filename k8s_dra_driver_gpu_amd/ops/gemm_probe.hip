@@ -1425,3 +1425,385 @@ gemm_fp8_256x256_g16_kernel(const unsigned char* A, const unsigned char* Bt, flo
                             int M, int N, int K) {
     gemm_fp8_256x256_body<16>(A, Bt, C, M, N, K);
 }
+
+// ---------------------------------------------------------------------------
+// MX-fp4 (OCP e2m1, scale=1) GEMM — mfma_scale_f32_32x32x64_f8f6f4, the
+// ~10 PF dense headline shape (ubench ceiling measured 9074 TF,
+// gpurun_out/r2s17). Elements pack two per byte, so HBM+LDS bytes per FLOP
+// are HALF the fp8 kernel's. 256x256 tile, 8 waves of 64x128 as 2x4
+// fragments of 32x32 (C/D: 16 f32 per lane, row = (reg&3)+8*(reg>>2)+
+// 4*(lane>>5), col = lane&31). BK=256 (4 MFMA K-depths per staged step),
+// 2 x 64 KiB LDS double buffer with dual barrier, grouped tile swizzle.
+// ---------------------------------------------------------------------------
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+typedef int i32x4 __attribute__((ext_vector_type(4)));
+
+template <int GROUP>
+__device__ __forceinline__ void gemm_fp4_256x256_body(
+    const unsigned char* __restrict__ A,   // [M][K/2] packed e2m1 (even k = low nibble)
+    const unsigned char* __restrict__ Bt,  // [N][K/2] packed e2m1
+    float* __restrict__ C,                 // [M][N] f32
+    int M, int N, int K) {
+    constexpr int TBM = 256, TBN = 256, BK = 256, NT = 512;
+    constexpr int AI = 2, BJ = 4;  // wave tile 64x128 in 32x32 fragments
+    constexpr int RB = BK / 2;     // tile row bytes (128)
+    __shared__ unsigned char lds[2 * (TBM + TBN) * RB];
+    const int HALF = (TBM + TBN) * RB;
+    auto ldsA = [&](int buf) -> unsigned char* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> unsigned char* { return lds + buf * HALF + TBM * RB; };
+
+    const int tiles_n = (N + TBN - 1) / TBN;
+    int tile_m, tile_n;
+    if (GROUP > 1) {
+        const int tiles_m = (M + TBM - 1) / TBM;
+        const int per_group = GROUP * tiles_n;
+        const int gid = blockIdx.x / per_group;
+        const int first_m = gid * GROUP;
+        const int gsz = min(GROUP, tiles_m - first_m);
+        tile_m = first_m + (blockIdx.x % per_group) % gsz;
+        tile_n = (blockIdx.x % per_group) / gsz;
+    } else {
+        tile_m = blockIdx.x / tiles_n;
+        tile_n = blockIdx.x % tiles_n;
+    }
+    const int m0 = tile_m * TBM;
+    const int n0 = tile_n * TBN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid >> 1) * 64;   // 4 wave rows
+    const int wc = (wid & 1) * 128;   // 2 wave cols
+
+    constexpr int ACH = TBM * (RB / 16);  // 16B chunks per A tile (2048)
+    constexpr int APH = ACH / NT;         // 4
+    auto stage = [&](int buf, int k0) {
+        unsigned char* la = ldsA(buf);
+        unsigned char* lb = ldsB(buf);
+        const size_t kb0 = (size_t)k0 / 2;
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (RB / 16);
+            int c = chunk % (RB / 16);
+            const unsigned char* ga = &A[(size_t)(m0 + r) * (K / 2) + kb0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+            const unsigned char* gb = &Bt[(size_t)(n0 + r) * (K / 2) + kb0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+    };
+
+    f32x16 acc[AI][BJ] = {};
+    const int kgrp = lane >> 5;       // 0..1: this lane's 32-k half
+    const int ln31 = lane & 31;
+    const int steps = K / BK;
+
+    stage(0, 0);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s & 1;
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 1 < steps) stage(buf ^ 1, (s + 1) * BK);
+
+        const unsigned char* la = ldsA(buf);
+        const unsigned char* lb = ldsB(buf);
+#pragma unroll
+        for (int d = 0; d < BK / 64; ++d) {  // 4 MFMA K-depths per step
+            const int kb = d * 32 + kgrp * 16;  // byte offset of this lane's 16B
+            union { i32x8 v; unsigned char b[32]; } af[AI], bf[BJ];
+#pragma unroll
+            for (int i = 0; i < AI; ++i) {
+                const int ar = wr + i * 32 + ln31;
+                *(i32x4*)&af[i].b[0] = *(const i32x4*)&la[ar * RB + kb];
+            }
+#pragma unroll
+            for (int j = 0; j < BJ; ++j) {
+                const int bc = wc + j * 32 + ln31;
+                *(i32x4*)&bf[j].b[0] = *(const i32x4*)&lb[bc * RB + kb];
+            }
+#pragma unroll
+            for (int i = 0; i < AI; ++i)
+#pragma unroll
+                for (int j = 0; j < BJ; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+                        af[i].v, bf[j].v, acc[i][j], 4, 4, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+        }
+        __syncthreads();
+    }
+
+#pragma unroll
+    for (int i = 0; i < AI; ++i)
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                int row = m0 + wr + i * 32 + (r & 3) + 8 * (r >> 2) + 4 * kgrp;
+                int col = n0 + wc + j * 32 + ln31;
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp4_256x256_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                        int M, int N, int K) {
+    gemm_fp4_256x256_body<1>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp4_256x256_g16_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                            int M, int N, int K) {
+    gemm_fp4_256x256_body<16>(A, Bt, C, M, N, K);
+}
+
+// fp4 3-buffer counted-vmcnt variant: BK=128 (2 MFMA K-depths per step),
+// 3 x 32 KiB LDS, single barrier per step (one tile in flight) — measures
+// whether the 2-buffer dual-barrier serialization is the remaining cost.
+template <int GROUP>
+__device__ __forceinline__ void gemm_fp4_3buf_body(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+    constexpr int TBM = 256, TBN = 256, BK = 128, NT = 512;
+    constexpr int AI = 2, BJ = 4;
+    constexpr int RB = BK / 2;  // 64 B rows
+    __shared__ unsigned char lds[3 * (TBM + TBN) * RB];
+    const int HALF = (TBM + TBN) * RB;
+    auto ldsA = [&](int buf) -> unsigned char* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> unsigned char* { return lds + buf * HALF + TBM * RB; };
+
+    const int tiles_n = (N + TBN - 1) / TBN;
+    int tile_m, tile_n;
+    if (GROUP > 1) {
+        const int tiles_m = (M + TBM - 1) / TBM;
+        const int per_group = GROUP * tiles_n;
+        const int gid = blockIdx.x / per_group;
+        const int first_m = gid * GROUP;
+        const int gsz = min(GROUP, tiles_m - first_m);
+        tile_m = first_m + (blockIdx.x % per_group) % gsz;
+        tile_n = (blockIdx.x % per_group) / gsz;
+    } else {
+        tile_m = blockIdx.x / tiles_n;
+        tile_n = blockIdx.x % tiles_n;
+    }
+    const int m0 = tile_m * TBM;
+    const int n0 = tile_n * TBN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid >> 1) * 64;
+    const int wc = (wid & 1) * 128;
+
+    constexpr int ACH = TBM * (RB / 16);  // 1024 chunks
+    constexpr int APH = ACH / NT;         // 2
+    constexpr int S = 2 * APH;            // A + B glds per thread per stage
+    auto stage = [&](int buf, int k0) {
+        unsigned char* la = ldsA(buf);
+        unsigned char* lb = ldsB(buf);
+        const size_t kb0 = (size_t)k0 / 2;
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (RB / 16);
+            int c = chunk % (RB / 16);
+            const unsigned char* ga = &A[(size_t)(m0 + r) * (K / 2) + kb0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+            const unsigned char* gb = &Bt[(size_t)(n0 + r) * (K / 2) + kb0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+    };
+
+    f32x16 acc[AI][BJ] = {};
+    const int kgrp = lane >> 5;
+    const int ln31 = lane & 31;
+    const int steps = K / BK;
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s % 3;
+        if (s + 1 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 2 < steps) stage((s + 2) % 3, (s + 2) * BK);
+
+        const unsigned char* la = ldsA(buf);
+        const unsigned char* lb = ldsB(buf);
+#pragma unroll
+        for (int d = 0; d < BK / 64; ++d) {
+            const int kb = d * 32 + kgrp * 16;
+            union { i32x8 v; unsigned char b[32]; } af[AI], bf[BJ];
+#pragma unroll
+            for (int i = 0; i < AI; ++i) {
+                const int ar = wr + i * 32 + ln31;
+                *(i32x4*)&af[i].b[0] = *(const i32x4*)&la[ar * RB + kb];
+            }
+#pragma unroll
+            for (int j = 0; j < BJ; ++j) {
+                const int bc = wc + j * 32 + ln31;
+                *(i32x4*)&bf[j].b[0] = *(const i32x4*)&lb[bc * RB + kb];
+            }
+#pragma unroll
+            for (int i = 0; i < AI; ++i)
+#pragma unroll
+                for (int j = 0; j < BJ; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+                        af[i].v, bf[j].v, acc[i][j], 4, 4, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < AI; ++i)
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                int row = m0 + wr + i * 32 + (r & 3) + 8 * (r >> 2) + 4 * kgrp;
+                int col = n0 + wc + j * 32 + ln31;
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp4_3buf_g16_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                         int M, int N, int K) {
+    gemm_fp4_3buf_body<16>(A, Bt, C, M, N, K);
+}
+
+// fp8 on the 32x32x64 shape (FMT=0), same 3-buffer counted structure as
+// the fp4 winner: 1 B/elem so rows are BK bytes; per lane 32 k = 32 B
+// fragments. 3 x 32 KiB LDS at BK=64.
+template <int GROUP>
+__device__ __forceinline__ void gemm_fp8k64_3buf_body(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+    constexpr int TBM = 256, TBN = 256, BK = 64, NT = 512;
+    constexpr int AI = 2, BJ = 4;
+    constexpr int RB = BK;  // 64 B rows (1 B/elem)
+    __shared__ unsigned char lds[3 * (TBM + TBN) * RB];
+    const int HALF = (TBM + TBN) * RB;
+    auto ldsA = [&](int buf) -> unsigned char* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> unsigned char* { return lds + buf * HALF + TBM * RB; };
+
+    const int tiles_n = (N + TBN - 1) / TBN;
+    int tile_m, tile_n;
+    if (GROUP > 1) {
+        const int tiles_m = (M + TBM - 1) / TBM;
+        const int per_group = GROUP * tiles_n;
+        const int gid = blockIdx.x / per_group;
+        const int first_m = gid * GROUP;
+        const int gsz = min(GROUP, tiles_m - first_m);
+        tile_m = first_m + (blockIdx.x % per_group) % gsz;
+        tile_n = (blockIdx.x % per_group) / gsz;
+    } else {
+        tile_m = blockIdx.x / tiles_n;
+        tile_n = blockIdx.x % tiles_n;
+    }
+    const int m0 = tile_m * TBM;
+    const int n0 = tile_n * TBN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid >> 1) * 64;
+    const int wc = (wid & 1) * 128;
+
+    constexpr int ACH = TBM * (RB / 16);  // 1024 chunks
+    constexpr int APH = ACH / NT;         // 2
+    constexpr int S = 2 * APH;            // A + B glds per thread per stage
+    auto stage = [&](int buf, int k0) {
+        unsigned char* la = ldsA(buf);
+        unsigned char* lb = ldsB(buf);
+        const size_t kb0 = (size_t)k0;
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (RB / 16);
+            int c = chunk % (RB / 16);
+            const unsigned char* ga = &A[(size_t)(m0 + r) * K + kb0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+            const unsigned char* gb = &Bt[(size_t)(n0 + r) * K + kb0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+    };
+
+    f32x16 acc[AI][BJ] = {};
+    const int kgrp = lane >> 5;
+    const int ln31 = lane & 31;
+    const int steps = K / BK;
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s % 3;
+        if (s + 1 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 2 < steps) stage((s + 2) % 3, (s + 2) * BK);
+
+        const unsigned char* la = ldsA(buf);
+        const unsigned char* lb = ldsB(buf);
+        {
+            const int kb = kgrp * 32;  // one 64-deep MFMA per step; 32 B/lane
+            i32x8 af[AI], bf[BJ];
+#pragma unroll
+            for (int i = 0; i < AI; ++i) {
+                const int ar = wr + i * 32 + ln31;
+                af[i] = *(const i32x8*)&la[ar * RB + kb];
+            }
+#pragma unroll
+            for (int j = 0; j < BJ; ++j) {
+                const int bc = wc + j * 32 + ln31;
+                bf[j] = *(const i32x8*)&lb[bc * RB + kb];
+            }
+#pragma unroll
+            for (int i = 0; i < AI; ++i)
+#pragma unroll
+                for (int j = 0; j < BJ; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+                        af[i], bf[j], acc[i][j], 0, 0, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < AI; ++i)
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                int row = m0 + wr + i * 32 + (r & 3) + 8 * (r >> 2) + 4 * kgrp;
+                int col = n0 + wc + j * 32 + ln31;
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8k64_3buf_g16_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                            int M, int N, int K) {
+    gemm_fp8k64_3buf_body<16>(A, Bt, C, M, N, K);
+}

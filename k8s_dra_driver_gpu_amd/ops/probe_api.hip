@@ -41,6 +41,8 @@ extern "C" __global__ void hbm_copy_kernel(float4v*, const float4v*, long);
 extern "C" __global__ void hbm_block_sum_kernel(const float*, float*, long);
 extern "C" __global__ void mfma_bf16_loop_kernel(const short*, float*, int);
 extern "C" __global__ void mfma_fp8_loop_kernel(const int*, float*, int);
+extern "C" __global__ void mfma_fp4_loop_kernel(const int*, float*, int);
+extern "C" __global__ void mfma_fp4_tile_gemm_kernel(const unsigned char*, const unsigned char*, float*, int);
 extern "C" __global__ void mfma_bf16_tile_gemm_kernel(const short*, const short*, float*, int);
 extern "C" __global__ void mfma_fp8_tile_gemm_kernel(const unsigned char*, const unsigned char*, float*, int, int);
 extern "C" __global__ void p2p_read_kernel(float4v*, const float4v*, long);
@@ -66,6 +68,10 @@ extern "C" __global__ void gemm_fp8_256_g16_kernel(const unsigned char*, const u
 extern "C" __global__ void gemm_fp8_256_g32_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp8_256x256_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp8_256x256_g16_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp4_256x256_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp4_256x256_g16_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp4_3buf_g16_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8k64_3buf_g16_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256x256_d2_g4_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256x256_d2_g8_kernel(const short*, const short*, float*, int, int, int);
 
@@ -580,6 +586,10 @@ static inline void (*fp8_kern_for(int v))(const unsigned char*, const unsigned c
         case 232: return gemm_fp8_256_g32_kernel;
         case 3:   return gemm_fp8_256x256_kernel;
         case 316: return gemm_fp8_256x256_g16_kernel;
+        case 4:   return gemm_fp4_256x256_kernel;
+        case 416: return gemm_fp4_256x256_g16_kernel;
+        case 436: return gemm_fp4_3buf_g16_kernel;
+        case 336: return gemm_fp8k64_3buf_g16_kernel;
         default: return gemm_fp8_128_kernel;
     }
 }
@@ -587,20 +597,21 @@ static inline void fp8_geom_for(int v, int* tm, int* tn, int* threads) {
     if (v == 2 || v == 24 || v == 28 || v == 216 || v == 232) {
         *tm = 256; *tn = 128; *threads = 512; return;
     }
-    if (v == 3 || v == 316) { *tm = 256; *tn = 256; *threads = 512; return; }
+    if (v == 3 || v == 316 || v == 336 || v == 4 || v == 416 || v == 436) { *tm = 256; *tn = 256; *threads = 512; return; }
     *tm = 128; *tn = 128; *threads = 256;
 }
 
 double fp_gemm_fp8_tflops_ex(int dev, int size, int iters, int variant) {
     CHK(hipSetDevice(dev));
     int M = size, N = size, K = size;
+    const int fp4 = (variant == 4 || variant == 416 || variant == 436);
     unsigned char *A, *Bt;
     float* C;
-    CHK(hipMalloc(&A, (size_t)M * K));
-    CHK(hipMalloc(&Bt, (size_t)N * K));
+    CHK(hipMalloc(&A, (size_t)M * K / (fp4 ? 2 : 1)));
+    CHK(hipMalloc(&Bt, (size_t)N * K / (fp4 ? 2 : 1)));
     CHK(hipMalloc(&C, (size_t)M * N * sizeof(float)));
-    CHK(hipMemset(A, 0x38, (size_t)M * K));   // e4m3 1.0
-    CHK(hipMemset(Bt, 0x30, (size_t)N * K));  // e4m3 0.5
+    CHK(hipMemset(A, fp4 ? 0x22 : 0x38, (size_t)M * K / (fp4 ? 2 : 1)));   // 1.0s
+    CHK(hipMemset(Bt, fp4 ? 0x11 : 0x30, (size_t)N * K / (fp4 ? 2 : 1)));  // 0.5s
     int tm, tn, threads;
     fp8_geom_for(variant, &tm, &tn, &threads);
     dim3 grid((M / tm) * (N / tn));
@@ -623,6 +634,55 @@ double fp_gemm_fp8_tflops_ex(int dev, int size, int iters, int variant) {
     hipEventDestroy(t0);
     hipEventDestroy(t1);
     return 2.0 * M * (double)N * K * iters / (ms * 1e9);
+}
+
+double fp_mfma_fp4_tflops(int dev, int inner_iters, int launches) {
+    CHK(hipSetDevice(dev));
+    int* seed;
+    float* sink;
+    CHK(hipMalloc(&seed, 1024 * sizeof(int)));
+    CHK(hipMalloc(&sink, sizeof(float)));
+    CHK(hipMemset(seed, 0x11, 1024 * sizeof(int)));
+    hipEvent_t t0, t1;
+    CHK(hipEventCreate(&t0));
+    CHK(hipEventCreate(&t1));
+    hipLaunchKernelGGL(mfma_fp4_loop_kernel, dim3(PROBE_GRID), dim3(PROBE_BLOCK), 0, 0, seed, sink, inner_iters);
+    CHK(hipGetLastError());
+    CHK(hipDeviceSynchronize());
+    CHK(hipEventRecord(t0));
+    for (int l = 0; l < launches; ++l)
+        hipLaunchKernelGGL(mfma_fp4_loop_kernel, dim3(PROBE_GRID), dim3(PROBE_BLOCK), 0, 0, seed, sink, inner_iters);
+    CHK(hipEventRecord(t1));
+    CHK(hipEventSynchronize(t1));
+    double ms = time_kernel_ms(t0, t1);
+    double waves = (double)PROBE_GRID * PROBE_BLOCK / 64.0;
+    // 4 MFMA per inner iter, each 2*32*32*64 FLOP
+    double flops = waves * 4.0 * inner_iters * launches * 2.0 * 32 * 32 * 64;
+    hipFree(seed);
+    hipFree(sink);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    return flops / (ms * 1e9);
+}
+
+int fp_mfma_fp4_tile_gemm_host(int dev, const unsigned char* A, const unsigned char* B,
+                               float* D, int K) {
+    CHKI(hipSetDevice(dev));
+    unsigned char *dA, *dB;
+    float* dD;
+    CHKI(hipMalloc(&dA, (size_t)32 * K / 2));
+    CHKI(hipMalloc(&dB, (size_t)K / 2 * 32));
+    CHKI(hipMalloc(&dD, 32 * 32 * sizeof(float)));
+    CHKI(hipMemcpy(dA, A, (size_t)32 * K / 2, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dB, B, (size_t)K / 2 * 32, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(mfma_fp4_tile_gemm_kernel, dim3(1), dim3(64), 0, 0, dA, dB, dD, K);
+    CHKI(hipGetLastError());
+    CHKI(hipDeviceSynchronize());
+    CHKI(hipMemcpy(D, dD, 32 * 32 * sizeof(float), hipMemcpyDeviceToHost));
+    hipFree(dA);
+    hipFree(dB);
+    hipFree(dD);
+    return 0;
 }
 
 double fp_mfma_fp8_tflops(int dev, int inner_iters, int launches) {
@@ -657,13 +717,14 @@ double fp_mfma_fp8_tflops(int dev, int inner_iters, int launches) {
 int fp_gemm_fp8_host_ex(int dev, const unsigned char* A, const unsigned char* Bt,
                         float* C, int M, int N, int K, int variant) {
     CHKI(hipSetDevice(dev));
+    const int den = (variant == 4 || variant == 416 || variant == 436) ? 2 : 1;  // fp4: packed bytes
     unsigned char *dA, *dB;
     float* dC;
-    CHKI(hipMalloc(&dA, (size_t)M * K));
-    CHKI(hipMalloc(&dB, (size_t)N * K));
+    CHKI(hipMalloc(&dA, (size_t)M * K / den));
+    CHKI(hipMalloc(&dB, (size_t)N * K / den));
     CHKI(hipMalloc(&dC, (size_t)M * N * sizeof(float)));
-    CHKI(hipMemcpy(dA, A, (size_t)M * K, hipMemcpyHostToDevice));
-    CHKI(hipMemcpy(dB, Bt, (size_t)N * K, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dA, A, (size_t)M * K / den, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dB, Bt, (size_t)N * K / den, hipMemcpyHostToDevice));
     int tm, tn, threads;
     fp8_geom_for(variant, &tm, &tn, &threads);
     dim3 grid(((M + tm - 1) / tm) * ((N + tn - 1) / tn));
