@@ -59,6 +59,8 @@ class DeviceLib:
         self.backend = backend or SysfsBackend()
         self._lock = threading.RLock()
         self._cache: Optional[Tuple[List[GpuInfo], List[PartitionInfo], GpuTopology]] = None
+        # accelerator profile table (amd-smi), read once on first use
+        self._acc_profiles: Optional[dict] = None
 
     # -- enumeration (cached) ---------------------------------------------
 
@@ -101,6 +103,38 @@ class DeviceLib:
 
     def supported_memory_modes(self, gpu: GpuInfo) -> List[str]:
         return self.backend.available_memory_partitions(gpu.minor)
+
+    def accelerator_profile_error(
+        self, gpu: GpuInfo, mode: str, memory_mode: str = ""
+    ) -> Optional[str]:
+        """Cross-check a requested partition switch against the platform's
+        accelerator profile table (`amd-smi partition --accelerator`, the
+        inspectMigProfilesAndPlacements analog). Returns an error string
+        when the table explicitly rejects the combination; None when OK or
+        unverifiable (CLI absent — mock trees, minimal containers)."""
+        if self.backend.sysfs_root != "/sys":
+            return None  # mock: sysfs availability is the only source
+        if self._acc_profiles is None:
+            from .acceleratorprofiles import read_accelerator_profiles
+
+            self._acc_profiles = read_accelerator_profiles() or {}
+        profiles = self._acc_profiles.get(gpu.index)
+        if not profiles:
+            return None
+        from .acceleratorprofiles import profile_for_mode
+
+        prof = profile_for_mode(profiles, mode)
+        if prof is None:
+            return (
+                f"mode {mode} not in the platform's accelerator profile table "
+                f"({[p.type for p in profiles]})"
+            )
+        if memory_mode and prof.memory_caps and memory_mode.upper() not in prof.memory_caps:
+            return (
+                f"memory mode {memory_mode} not supported by the {mode} "
+                f"profile (caps: {prof.memory_caps})"
+            )
+        return None
 
     def possible_partitions(self, gpu: GpuInfo) -> List[PartitionSpec]:
         """All partition placements this GPU could host (for ResourceSlice
@@ -149,6 +183,9 @@ class DeviceLib:
                         f"cannot switch to {spec.compute_mode} while partitioned"
                     )
                 self._assert_quiesced(gpu)
+                perr = self.accelerator_profile_error(gpu, spec.compute_mode, memory_mode)
+                if perr:
+                    raise DeviceError(f"partition profile check: {perr}")
                 if memory_mode and memory_mode != gpu.memory_partition:
                     min_parts = MEMORY_MODE_MIN_PARTITIONS.get(memory_mode, 1)
                     if nparts < min_parts:

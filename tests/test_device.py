@@ -289,3 +289,58 @@ class TestPartitionWriteFallbacks:
         with pytest.raises(SysfsError, match="busy"):
             b.set_compute_partition(0, "CPX")
         assert calls == []
+
+
+class TestAcceleratorProfiles:
+    """amd-smi accelerator-partition profile parsing (the MIG
+    profile/placement inspection analog, nvlib.go:1202-1277) against REAL
+    CLI output captured on an MI355X (tests/fixtures/)."""
+
+    def _fixture(self):
+        p = os.path.join(os.path.dirname(__file__), "fixtures",
+                         "amd_smi_partition_accelerator.txt")
+        return open(p).read()
+
+    def test_parse_real_capture(self):
+        from k8s_dra_driver_gpu_amd.device.acceleratorprofiles import (
+            parse_accelerator_profiles,
+        )
+
+        profs = parse_accelerator_profiles(self._fixture())
+        assert 0 in profs
+        by_type = {p.type: p for p in profs[0]}
+        assert by_type["SPX"].current is True
+        assert by_type["SPX"].num_partitions == 1
+        assert by_type["SPX"].resources["XCC"] == (8, 1)
+        assert by_type["SPX"].resources["JPEG"] == (40, 1)
+        assert by_type["DPX"].num_partitions == 2
+        assert by_type["DPX"].current is False
+        assert by_type["QPX"].num_partitions == 4
+        assert by_type["SPX"].memory_caps == ["NPS1"]
+
+    def test_profile_for_mode(self):
+        from k8s_dra_driver_gpu_amd.device.acceleratorprofiles import (
+            parse_accelerator_profiles,
+            profile_for_mode,
+        )
+
+        profs = parse_accelerator_profiles(self._fixture())[0]
+        assert profile_for_mode(profs, "dpx").type == "DPX"
+        assert profile_for_mode(profs, "CPX") is None  # truncated capture
+
+    def test_missing_cli_returns_none(self):
+        from k8s_dra_driver_gpu_amd.device.acceleratorprofiles import (
+            read_accelerator_profiles,
+        )
+
+        assert read_accelerator_profiles(cli="/nonexistent/amd-smi") is None
+
+    def test_truncated_output_tolerated(self):
+        from k8s_dra_driver_gpu_amd.device.acceleratorprofiles import (
+            parse_accelerator_profiles,
+        )
+
+        text = self._fixture()
+        assert parse_accelerator_profiles(text[: len(text) // 2])
+        assert parse_accelerator_profiles("") == {}
+        assert parse_accelerator_profiles("garbage\nlines\n") == {}
