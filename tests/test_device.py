@@ -316,6 +316,7 @@ class TestAcceleratorProfiles:
         assert by_type["DPX"].num_partitions == 2
         assert by_type["DPX"].current is False
         assert by_type["QPX"].num_partitions == 4
+        assert by_type["CPX"].num_partitions == 8
         assert by_type["SPX"].memory_caps == ["NPS1"]
 
     def test_profile_for_mode(self):
@@ -326,7 +327,8 @@ class TestAcceleratorProfiles:
 
         profs = parse_accelerator_profiles(self._fixture())[0]
         assert profile_for_mode(profs, "dpx").type == "DPX"
-        assert profile_for_mode(profs, "CPX") is None  # truncated capture
+        assert profile_for_mode(profs, "cpx").num_partitions == 8
+        assert profile_for_mode(profs, "TPX") is None
 
     def test_missing_cli_returns_none(self):
         from k8s_dra_driver_gpu_amd.device.acceleratorprofiles import (
