@@ -471,6 +471,9 @@ class FabricProbeReport:
     gemm_bf16_tflops: float
     p2p_gbps: List[List[float]]  # [dst][src], -1 on self
     allreduce_gbps: float
+    # low-precision matrix-core floors (MX-scaled fp8/fp4; 0.0 = not run)
+    mfma_fp8_tflops: float = 0.0
+    mfma_fp4_tflops: float = 0.0
 
 
 def run_fabric_report(quick: bool = True) -> FabricProbeReport:
@@ -503,4 +506,6 @@ def run_fabric_report(quick: bool = True) -> FabricProbeReport:
         gemm_bf16_tflops=gemm_bf16_tflops(0, 2048 if quick else 4096, 5),
         p2p_gbps=p2p,
         allreduce_gbps=ar,
+        mfma_fp8_tflops=mfma_fp8_tflops(0, 1024, 5),
+        mfma_fp4_tflops=mfma_fp4_tflops(0, 1024, 5),
     )

@@ -523,3 +523,20 @@ class TestHttpWatchCache:
             inf.stop()
         finally:
             srv.stop()
+
+
+class TestHttpBookmarks:
+    def test_idle_http_watch_emits_bookmarks(self, client):
+        client.create("pods", {"apiVersion": "v1", "kind": "Pod",
+                               "metadata": {"name": "a", "namespace": "default"}})
+        _, rv = client.list_with_rv("pods")
+        w = client.watch("pods", resource_version=rv, allow_bookmarks=True)
+        deadline = time.monotonic() + 15
+        ev = None
+        while time.monotonic() < deadline:
+            ev = w.next(timeout=2.0)
+            if ev is not None:
+                break
+        assert ev is not None and ev.type == "BOOKMARK", ev
+        assert ev.object["metadata"]["resourceVersion"] == rv
+        w.stop()
