@@ -503,7 +503,10 @@ def run_fabric_report(quick: bool = True) -> FabricProbeReport:
         hbm_read_gbps=hbm_read_gbps(0, size, iters),
         hbm_write_gbps=hbm_write_gbps(0, size, iters),
         mfma_bf16_tflops=mfma_bf16_tflops(0, 1024, 10),
-        gemm_bf16_tflops=gemm_bf16_tflops(0, 2048 if quick else 4096, 5),
+        # quick mode uses the 128x128-tile kernel: a 2048^3 problem is only
+        # 64 workgroups for the 256x256 champion (256 CUs idle)
+        gemm_bf16_tflops=(gemm_bf16_tflops_ex(0, 2048, 5, 432) if quick
+                          else gemm_bf16_tflops(0, 8192, 5)),
         p2p_gbps=p2p,
         allreduce_gbps=ar,
         mfma_fp8_tflops=mfma_fp8_tflops(0, 1024, 5),
