@@ -206,7 +206,8 @@ def allreduce_pull_gbps(bytes_: int = 1 << 30, iters: int = 5) -> float:
 
 def gemm_bf16_tflops(dev: int = 0, size: int = 4096, iters: int = 10) -> float:
     """LDS-staged big-tile bf16 GEMM throughput (size^3 problem; dispatches
-    the measured champion: 256x256 tile, depth-2, register-hoisted)."""
+    the measured champion: 256x256 tile, depth-2, register-hoisted,
+    XOR-swizzled LDS)."""
     return _check(_load().fp_gemm_bf16_tflops(dev, size, iters), "gemm_bf16")
 
 
@@ -227,8 +228,8 @@ def gemm_bf16(a: np.ndarray, bt: np.ndarray, dev: int = 0, bk: int = 32) -> np.n
     # selector -> K-step depth: 232/264 = 32x32x16 tiling, 332/364 = the
     # 3-buffer pipelined kernel (counted vmcnt + raw barrier)
     kstep = {32: 32, 64: 64, 232: 32, 264: 64, 332: 32, 364: 64, 432: 32,
-             532: 32, 632: 32, 732: 32, 764: 64, 832: 32, 842: 32, 844: 32, 848: 32}[bk]
-    tile_m = {732: 256, 764: 256, 832: 256, 842: 256, 844: 256, 848: 256}.get(bk, 128)
+             532: 32, 632: 32, 732: 32, 764: 64, 832: 32, 842: 32, 844: 32, 848: 32, 852: 32}[bk]
+    tile_m = {732: 256, 764: 256, 832: 256, 842: 256, 844: 256, 848: 256, 852: 256}.get(bk, 128)
     assert K == K2 and M % tile_m == 0 and N % 128 == 0 and K % kstep == 0
     a_bf = _to_bf16_bits(np.ascontiguousarray(a, dtype=np.float32))
     b_bf = _to_bf16_bits(np.ascontiguousarray(bt, dtype=np.float32))
@@ -373,7 +374,7 @@ def mfma_fp4_tile_gemm(a: np.ndarray, b: np.ndarray, dev: int = 0) -> np.ndarray
 
 
 def gemm_fp4(a: np.ndarray, bt: np.ndarray, dev: int = 0,
-             variant: int = 436) -> np.ndarray:
+             variant: int = 446) -> np.ndarray:
     """C[M,N] = a[M,K] @ bt[N,K]^T on the MX-fp4 GEMM kernel (e2m1 packed
     two-per-byte, fp32 out). variant 4 = 256x256 2-buf, 416 = +G16 swizzle,
     436 = 3-buf counted pipeline G16."""
@@ -396,10 +397,10 @@ def gemm_fp4(a: np.ndarray, bt: np.ndarray, dev: int = 0,
 
 
 def gemm_fp4_tflops_ex(dev: int = 0, size: int = 4096, iters: int = 10,
-                       variant: int = 436) -> float:
-    """MX-fp4 GEMM throughput. Champion 436 (3-buf counted G16): measured
-    2649/3197 TF @4096^3/8192^3 vs the 9074 TF mfma ceiling
-    (gpurun_out/r2s17-20)."""
+                       variant: int = 446) -> float:
+    """MX-fp4 GEMM throughput. Champion 446 (3-buf counted G16 + XOR LDS
+    swizzle): measured 3060/3526 TF @4096^3/8192^3 vs the 9074 TF mfma
+    ceiling (gpurun_out/r2s17-20, r2s24)."""
     return _check(_load().fp_gemm_fp8_tflops_ex(dev, size, iters, variant),
                   "gemm_fp4_ex")
 
@@ -415,18 +416,18 @@ def mfma_fp8_tflops(dev: int = 0, inner_iters: int = 2048, launches: int = 20) -
 
 
 def gemm_fp8_tflops_ex(dev: int = 0, size: int = 4096, iters: int = 10,
-                       variant: int = 316) -> float:
+                       variant: int = 326) -> float:
     """MX-fp8 (e4m3, scale=1) GEMM throughput. Variants: 1 = 128x128;
     2/24/28/216/232 = 256x128 3-buf (suffix = tile-group swizzle);
-    3/316 = 256x256 2-buf dual-barrier (champion: 316 = G16 swizzle,
-    measured 1469/1625 TF @4096^3/8192^3 vs the 4780 TF mfma_scale
-    ceiling — gpurun_out/r2s15)."""
+    3/316 = 256x256 2-buf dual-barrier + G16; champion 326 = 316 with the
+    XOR LDS bank swizzle: 1836/2003 TF @4096^3/8192^3 vs the 4780 TF
+    mfma_scale ceiling (gpurun_out/r2s15, r2s24)."""
     return _check(_load().fp_gemm_fp8_tflops_ex(dev, size, iters, variant),
                   "gemm_fp8_ex")
 
 
 def gemm_fp8(a: np.ndarray, bt: np.ndarray, dev: int = 0,
-             variant: int = 316) -> np.ndarray:
+             variant: int = 326) -> np.ndarray:
     """C[M,N] = a[M,K] @ bt[N,K]^T on the MX-fp8 GEMM kernel (e4m3 in,
     fp32 out); inputs are float32, quantized to e4m3 exactly as consumed."""
     M, K = a.shape

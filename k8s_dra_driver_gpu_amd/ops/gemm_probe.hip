@@ -982,7 +982,7 @@ gemm_bf16_256_bk64_kernel(const short* A, const short* Bt, float* C, int M, int 
 // Depth-2 variant of the generic big-tile body: 4 LDS buffers, TWO tiles in
 // flight across each barrier (pipe2-style counted vmcnt). For 256x256 this
 // is 4 x 32 KiB = 128 KiB LDS (fits the 160 KiB CU budget at 1 WG/CU).
-template <int TBM, int TBN, int TWR, int TWC, int BK, int GROUP = 1>
+template <int TBM, int TBN, int TWR, int TWC, int BK, int GROUP = 1, int SWZ = 0>
 __device__ __forceinline__ void gemm_big2_body(
     const short* __restrict__ A, const short* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
@@ -1034,6 +1034,10 @@ __device__ __forceinline__ void gemm_big2_body(
             int chunk = phase * NT + tid;
             int r = chunk / (BK / 8);
             int c = chunk % (BK / 8);
+            // SWZ: LDS slot s holds global chunk (r, c ^ (r & CPR-1)) —
+            // rows land on different banks so the 16-lane fragment reads
+            // (row stride = one bank group) stop conflicting
+            if (SWZ) c ^= r & (BK / 8 - 1);
             const short* ga = &A[(size_t)(m0 + r) * K + k0 + c * 8];
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) void*)ga,
@@ -1045,6 +1049,7 @@ __device__ __forceinline__ void gemm_big2_body(
             int chunk = phase * NT + tid;
             int r = chunk / (BK / 8);
             int c = chunk % (BK / 8);
+            if (SWZ) c ^= r & (BK / 8 - 1);
             const short* gb = &Bt[(size_t)(n0 + r) * K + k0 + c * 8];
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) void*)gb,
@@ -1080,12 +1085,16 @@ __device__ __forceinline__ void gemm_big2_body(
 #pragma unroll
             for (int i = 0; i < AI; ++i) {
                 const int ar = wr + i * 16 + (lane & 15);
-                af[i] = *(const bf16x8*)&la[ar * BK + ks * 32 + kg];
+                int ch = ks * 4 + (lane >> 4);  // 16B chunk within the row
+                if (SWZ) ch ^= ar & (BK / 8 - 1);
+                af[i] = *(const bf16x8*)&la[ar * BK + ch * 8];
             }
 #pragma unroll
             for (int j = 0; j < BJ; ++j) {
                 const int bc = wc + j * 16 + (lane & 15);
-                bf[j] = *(const bf16x8*)&lb[bc * BK + ks * 32 + kg];
+                int ch = ks * 4 + (lane >> 4);
+                if (SWZ) ch ^= bc & (BK / 8 - 1);
+                bf[j] = *(const bf16x8*)&lb[bc * BK + ch * 8];
             }
 #pragma unroll
             for (int i = 0; i < AI; ++i)
@@ -1111,6 +1120,11 @@ __device__ __forceinline__ void gemm_big2_body(
 extern "C" __global__ void __launch_bounds__(512)
 gemm_bf16_256x256_d2_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
     gemm_big2_body<256, 256, 4, 2, 32>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_bf16_256x256_d2_swz_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
+    gemm_big2_body<256, 256, 4, 2, 32, 1, 1>(A, Bt, C, M, N, K);
 }
 
 extern "C" __global__ void __launch_bounds__(512)
@@ -1307,7 +1321,7 @@ gemm_fp8_256_g32_kernel(const unsigned char* A, const unsigned char* Bt, float* 
 
 // 256x256 fp8, 2 x 64 KiB LDS double buffer (dual barrier per step since a
 // staged buffer is immediately reused), 8 waves of 64x128 (AI=4, BJ=8)
-template <int GROUP>
+template <int GROUP, int SWZ = 0>
 __device__ __forceinline__ void gemm_fp8_256x256_body(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
@@ -1351,6 +1365,7 @@ __device__ __forceinline__ void gemm_fp8_256x256_body(
             int chunk = phase * NT + tid;
             int r = chunk / (BK / 16);
             int c = chunk % (BK / 16);
+            if (SWZ) c = ((c >> 1) ^ (r & 3)) * 2 + (c & 1);  // pair swizzle
             const unsigned char* ga = &A[(size_t)(m0 + r) * K + k0 + c * 16];
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) void*)ga,
@@ -1365,7 +1380,6 @@ __device__ __forceinline__ void gemm_fp8_256x256_body(
     };
 
     f32x4 acc[AI][BJ] = {};
-    const int kg = (lane >> 4) * 32;
     const int steps = K / BK;
 
     stage(0, 0);
@@ -1385,12 +1399,16 @@ __device__ __forceinline__ void gemm_fp8_256x256_body(
 #pragma unroll
         for (int i = 0; i < AI; ++i) {
             const int ar = wr + i * 16 + (lane & 15);
-            af[i] = *(const i32x8*)&la[ar * BK + kg];
+            int p = lane >> 4;  // 32B pair index of this lane's k-slice
+            if (SWZ) p ^= ar & 3;
+            af[i] = *(const i32x8*)&la[ar * BK + p * 32];
         }
 #pragma unroll
         for (int j = 0; j < BJ; ++j) {
             const int bc = wc + j * 16 + (lane & 15);
-            bf[j] = *(const i32x8*)&lb[bc * BK + kg];
+            int p = lane >> 4;
+            if (SWZ) p ^= bc & 3;
+            bf[j] = *(const i32x8*)&lb[bc * BK + p * 32];
         }
 #pragma unroll
         for (int i = 0; i < AI; ++i)
@@ -1424,6 +1442,12 @@ extern "C" __global__ void __launch_bounds__(512)
 gemm_fp8_256x256_g16_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
                             int M, int N, int K) {
     gemm_fp8_256x256_body<16>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_256x256_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                                int M, int N, int K) {
+    gemm_fp8_256x256_body<16, 1>(A, Bt, C, M, N, K);
 }
 
 // ---------------------------------------------------------------------------
@@ -1566,7 +1590,7 @@ gemm_fp4_256x256_g16_kernel(const unsigned char* A, const unsigned char* Bt, flo
 // fp4 3-buffer counted-vmcnt variant: BK=128 (2 MFMA K-depths per step),
 // 3 x 32 KiB LDS, single barrier per step (one tile in flight) — measures
 // whether the 2-buffer dual-barrier serialization is the remaining cost.
-template <int GROUP>
+template <int GROUP, int SWZ = 0>
 __device__ __forceinline__ void gemm_fp4_3buf_body(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
@@ -1613,6 +1637,7 @@ __device__ __forceinline__ void gemm_fp4_3buf_body(
             int chunk = phase * NT + tid;
             int r = chunk / (RB / 16);
             int c = chunk % (RB / 16);
+            if (SWZ) c ^= r & (RB / 16 - 1);
             const unsigned char* ga = &A[(size_t)(m0 + r) * (K / 2) + kb0 + c * 16];
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) void*)ga,
@@ -1647,17 +1672,20 @@ __device__ __forceinline__ void gemm_fp4_3buf_body(
         const unsigned char* lb = ldsB(buf);
 #pragma unroll
         for (int d = 0; d < BK / 64; ++d) {
-            const int kb = d * 32 + kgrp * 16;
             union { i32x8 v; unsigned char b[32]; } af[AI], bf[BJ];
 #pragma unroll
             for (int i = 0; i < AI; ++i) {
                 const int ar = wr + i * 32 + ln31;
-                *(i32x4*)&af[i].b[0] = *(const i32x4*)&la[ar * RB + kb];
+                int ch = d * 2 + kgrp;
+                if (SWZ) ch ^= ar & (RB / 16 - 1);
+                *(i32x4*)&af[i].b[0] = *(const i32x4*)&la[ar * RB + ch * 16];
             }
 #pragma unroll
             for (int j = 0; j < BJ; ++j) {
                 const int bc = wc + j * 32 + ln31;
-                *(i32x4*)&bf[j].b[0] = *(const i32x4*)&lb[bc * RB + kb];
+                int ch = d * 2 + kgrp;
+                if (SWZ) ch ^= bc & (RB / 16 - 1);
+                *(i32x4*)&bf[j].b[0] = *(const i32x4*)&lb[bc * RB + ch * 16];
             }
 #pragma unroll
             for (int i = 0; i < AI; ++i)
@@ -1684,6 +1712,12 @@ extern "C" __global__ void __launch_bounds__(512)
 gemm_fp4_3buf_g16_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
                          int M, int N, int K) {
     gemm_fp4_3buf_body<16>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp4_3buf_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                             int M, int N, int K) {
+    gemm_fp4_3buf_body<16, 1>(A, Bt, C, M, N, K);
 }
 
 // fp8 on the 32x32x64 shape (FMT=0), same 3-buffer counted structure as

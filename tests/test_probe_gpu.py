@@ -258,8 +258,9 @@ class TestGemmProbe:
         rng = np.random.default_rng(17)
         for M, N, K, bk in ((512, 128, 160, 732), (256, 384, 224, 732),
                             (512, 512, 192, 832), (512, 256, 224, 842),
+                            (512, 256, 224, 852),
                             (1024, 1024, 1024, 732), (1024, 1024, 1024, 832),
-                            (1024, 1024, 1024, 842)):
+                            (1024, 1024, 1024, 842), (1024, 1024, 1024, 852)):
             a = rng.standard_normal((M, K), dtype=np.float32)
             bt = rng.standard_normal((N, K), dtype=np.float32)
             d = probe.gemm_bf16(a, bt, bk=bk)
@@ -343,7 +344,7 @@ class TestFp8Probe:
         bt = rng.standard_normal((N, K)).astype(np.float32)
         ref = (probe.fp8_e4m3_to_f32(probe.to_fp8_e4m3(a)).astype(np.float64)
                @ probe.fp8_e4m3_to_f32(probe.to_fp8_e4m3(bt)).astype(np.float64).T)
-        for v in (1, 2, 216, 3, 316):
+        for v in (1, 2, 216, 3, 316, 326):
             d = probe.gemm_fp8(a, bt, variant=v)
             err = np.abs(d - ref).max() / np.abs(ref).max()
             assert err < 1e-3, f"variant {v}: {err}"
@@ -401,7 +402,7 @@ class TestFp4Probe:
         bt = rng.standard_normal((N, K)).astype(np.float32)
         ref = (probe.fp4_e2m1_to_f32(probe.to_fp4_e2m1(a)).astype(np.float64)
                @ probe.fp4_e2m1_to_f32(probe.to_fp4_e2m1(bt)).astype(np.float64).T)
-        for v in (4, 416, 436):
+        for v in (4, 416, 436, 446):
             d = probe.gemm_fp4(a, bt, variant=v)
             err = np.abs(d - ref).max() / np.abs(ref).max()
             assert err < 1e-6, f"variant {v}: {err}"
