@@ -11,7 +11,10 @@ COPY deployments/ deployments/
 COPY bench.py __graft_entry__.py ./
 
 # build the CDNA4 probe library (gfx950) and the C++ fabric daemon
-RUN python -m k8s_dra_driver_gpu_amd.ops.build && make -C native -j4
+# (fabricd links OpenSSL for the optional mTLS peer-mesh mode)
+RUN apt-get update && apt-get install -y --no-install-recommends libssl-dev \
+    && rm -rf /var/lib/apt/lists/* \
+    && python -m k8s_dra_driver_gpu_amd.ops.build && make -C native -j4
 
 ENV PYTHONPATH=/opt/amd-dra-driver \
     FABRICD_PATH=/opt/amd-dra-driver/native/bin/fabricd \
