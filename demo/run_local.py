@@ -20,7 +20,8 @@ def main() -> int:
     cluster = LocalCluster(num_gpus=8).start()
     try:
         for spec in ("gpu-test1.yaml", "gpu-test2.yaml", "gpu-test3.yaml",
-                     "gpu-test-partitions.yaml", "gpu-test-extres.yaml"):
+                     "gpu-test5.yaml", "gpu-test-partitions.yaml",
+                     "gpu-test-extres.yaml"):
             print(f"=== {spec} ===")
             for ev in cluster.apply_yaml(os.path.join(SPECS, spec)):
                 print(" ", ev)
