@@ -71,7 +71,7 @@ def _free_port() -> int:
 class LocalCluster:
     def __init__(self, num_gpus: int = 8, node_name: str = "node-a",
                  work_dir: str = "", real_devices: bool = False,
-                 partitionable: bool = True):
+                 partitionable: bool = True, vfio: bool = False):
         self.client = FakeClient()
         self.node_name = node_name
         self.work_dir = work_dir or tempfile.mkdtemp(prefix="amddra-local-")
@@ -83,6 +83,7 @@ class LocalCluster:
             self.mock.setup()
             self.devicelib = DeviceLib(backend=self.mock.backend())
         self.partitionable = partitionable
+        self.vfio = vfio
         self.scheduler = SchedulerStub(self.client)
         self.controller: Optional[ComputeDomainController] = None
         self.gpu_driver: Optional[GpuDriver] = None
@@ -124,6 +125,7 @@ class LocalCluster:
             cdi=CdiHandler(cdi_root=os.path.join(self.work_dir, "cdi"), dev_root=dev_root),
             checkpoints=CheckpointManager(state_dir),
             state_dir=state_dir,
+            vfio=(self.mock.vfio_manager() if (self.vfio and self.mock) else None),
         )
         self.gpu_driver = GpuDriver(
             state=ds, claim_resolver=k8s_claim_resolver(self.client),
@@ -132,7 +134,8 @@ class LocalCluster:
         socks = self.gpu_driver.start(plugin_dir=os.path.join(self.work_dir, "plugin"))
         self.gpu_client = dra.DRAPluginClient(f"unix://{socks['dra']}")
         for sl in ResourceSliceGenerator(
-            self.devicelib, node_name=self.node_name, partitionable=self.partitionable
+            self.devicelib, node_name=self.node_name, partitionable=self.partitionable,
+            vfio=self.vfio
         ).generate():
             self.client.apply("resourceslices", sl)
         # CD plugin

@@ -119,6 +119,54 @@ class MockTree:
         pdir = os.path.join(self.sysfs_root, "class", "kfd", "kfd", "proc", str(pid))
         shutil.rmtree(pdir, ignore_errors=True)
 
+    def setup_vfio(self) -> None:
+        """Create the PCI-driver rebind surface for VFIO demos/tests:
+        per-GPU PCI device dirs with driver symlinks (amdgpu), IOMMU
+        groups, and /dev/vfio nodes — consumed by vfio_manager()."""
+        import glob as _glob
+
+        drivers = os.path.join(self.sysfs_root, "bus", "pci", "drivers")
+        for drv in ("amdgpu", "vfio-pci"):
+            os.makedirs(os.path.join(drivers, drv), exist_ok=True)
+        vdev = os.path.join(self.dev_root, "vfio")
+        os.makedirs(vdev, exist_ok=True)
+        if not os.path.exists(os.path.join(vdev, "vfio")):
+            open(os.path.join(vdev, "vfio"), "w").close()
+        for gpu in range(self.num_gpus):
+            pci = self.pci_addr(gpu)
+            devdir = os.path.join(self.sysfs_root, "bus", "pci", "devices", pci)
+            os.makedirs(devdir, exist_ok=True)
+            drv_link = os.path.join(devdir, "driver")
+            if not os.path.islink(drv_link):
+                os.symlink(os.path.join("..", "..", "drivers", "amdgpu"), drv_link)
+            group = str(40 + gpu)
+            gdir = os.path.join(self.sysfs_root, "kernel", "iommu_groups", group)
+            os.makedirs(gdir, exist_ok=True)
+            glink = os.path.join(devdir, "iommu_group")
+            if not os.path.islink(glink):
+                os.symlink(gdir, glink)
+            if not os.path.exists(os.path.join(vdev, group)):
+                open(os.path.join(vdev, group), "w").close()
+
+    def _vfio_rebind(self, pci: str, driver: str) -> None:
+        devdir = os.path.join(self.sysfs_root, "bus", "pci", "devices", pci)
+        link = os.path.join(devdir, "driver")
+        if os.path.islink(link):
+            os.unlink(link)
+        os.symlink(os.path.join("..", "..", "drivers", driver), link)
+
+    def vfio_manager(self):
+        """A VfioPciManager over this mock tree (functional rebind)."""
+        from ..plugin.vfio import VfioPciManager
+
+        self.setup_vfio()
+        return VfioPciManager(
+            sysfs_root=self.sysfs_root,
+            dev_root=self.dev_root,
+            busy_check=lambda pci: False,
+            rebind_hook=self._vfio_rebind,
+        )
+
     def set_memory_partition(self, gpu: int, mode: str) -> None:
         mode = mode.upper()
         avail = [m.strip() for m in self.profile.available_memory.split(",")]
