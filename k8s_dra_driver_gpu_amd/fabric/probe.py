@@ -416,18 +416,19 @@ def mfma_fp8_tflops(dev: int = 0, inner_iters: int = 2048, launches: int = 20) -
 
 
 def gemm_fp8_tflops_ex(dev: int = 0, size: int = 4096, iters: int = 10,
-                       variant: int = 326) -> float:
+                       variant: int = 346) -> float:
     """MX-fp8 (e4m3, scale=1) GEMM throughput. Variants: 1 = 128x128;
     2/24/28/216/232 = 256x128 3-buf (suffix = tile-group swizzle);
-    3/316 = 256x256 2-buf dual-barrier + G16; champion 326 = 316 with the
-    XOR LDS bank swizzle: 1836/2003 TF @4096^3/8192^3 vs the 4780 TF
-    mfma_scale ceiling (gpurun_out/r2s15, r2s24)."""
+    3/316 = 256x256 2-buf dual-barrier + G16; 326 adds the XOR LDS bank
+    swizzle; champion 346 further drops the provably-redundant trailing
+    barrier: 1824/1992-2003 TF @4096^3/8192^3 vs the 4780 TF mfma_scale
+    ceiling (gpurun_out/r2s15, r2s24, r2s26)."""
     return _check(_load().fp_gemm_fp8_tflops_ex(dev, size, iters, variant),
                   "gemm_fp8_ex")
 
 
 def gemm_fp8(a: np.ndarray, bt: np.ndarray, dev: int = 0,
-             variant: int = 326) -> np.ndarray:
+             variant: int = 346) -> np.ndarray:
     """C[M,N] = a[M,K] @ bt[N,K]^T on the MX-fp8 GEMM kernel (e4m3 in,
     fp32 out); inputs are float32, quantized to e4m3 exactly as consumed."""
     M, K = a.shape

@@ -1321,7 +1321,7 @@ gemm_fp8_256_g32_kernel(const unsigned char* A, const unsigned char* Bt, float* 
 
 // 256x256 fp8, 2 x 64 KiB LDS double buffer (dual barrier per step since a
 // staged buffer is immediately reused), 8 waves of 64x128 (AI=4, BJ=8)
-template <int GROUP, int SWZ = 0>
+template <int GROUP, int SWZ = 0, int TAILBAR = 1>
 __device__ __forceinline__ void gemm_fp8_256x256_body(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
@@ -1416,8 +1416,11 @@ __device__ __forceinline__ void gemm_fp8_256x256_body(
             for (int j = 0; j < BJ; ++j)
                 acc[i][j] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
                     af[i], bf[j], acc[i][j], 0, 0, 0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
-        // every wave done reading buf before the next stage overwrites it
-        __syncthreads();
+        // TAILBAR=0: the next iteration's top barrier already proves every
+        // wave finished this step's reads before stage(s+2) can overwrite
+        // the buffer (stages are only issued after a barrier) — measured
+        // variant; TAILBAR=1 keeps the conservative trailing barrier.
+        if (TAILBAR) __syncthreads();
     }
 
 #pragma unroll
@@ -1448,6 +1451,12 @@ extern "C" __global__ void __launch_bounds__(512)
 gemm_fp8_256x256_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
                                 int M, int N, int K) {
     gemm_fp8_256x256_body<16, 1>(A, Bt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_256x256_g16_swz_nb_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                                   int M, int N, int K) {
+    gemm_fp8_256x256_body<16, 1, 0>(A, Bt, C, M, N, K);
 }
 
 // ---------------------------------------------------------------------------

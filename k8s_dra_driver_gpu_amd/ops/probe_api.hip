@@ -73,6 +73,7 @@ extern "C" __global__ void gemm_fp4_256x256_g16_kernel(const unsigned char*, con
 extern "C" __global__ void gemm_fp4_3buf_g16_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp8k64_3buf_g16_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp8_256x256_g16_swz_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_256x256_g16_swz_nb_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp4_3buf_g16_swz_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256x256_d2_swz_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256x256_d2_g4_kernel(const short*, const short*, float*, int, int, int);
@@ -598,6 +599,7 @@ static inline void (*fp8_kern_for(int v))(const unsigned char*, const unsigned c
         case 446: return gemm_fp4_3buf_g16_swz_kernel;
         case 336: return gemm_fp8k64_3buf_g16_kernel;
         case 326: return gemm_fp8_256x256_g16_swz_kernel;
+        case 346: return gemm_fp8_256x256_g16_swz_nb_kernel;
         default: return gemm_fp8_128_kernel;
     }
 }
@@ -605,7 +607,7 @@ static inline void fp8_geom_for(int v, int* tm, int* tn, int* threads) {
     if (v == 2 || v == 24 || v == 28 || v == 216 || v == 232) {
         *tm = 256; *tn = 128; *threads = 512; return;
     }
-    if (v == 3 || v == 316 || v == 326 || v == 336 || v == 4 || v == 416 || v == 436 || v == 446) { *tm = 256; *tn = 256; *threads = 512; return; }
+    if (v == 3 || v == 316 || v == 326 || v == 336 || v == 346 || v == 4 || v == 416 || v == 436 || v == 446) { *tm = 256; *tn = 256; *threads = 512; return; }
     *tm = 128; *tn = 128; *threads = 256;
 }
 
