@@ -380,7 +380,8 @@ def gemm_fp4(a: np.ndarray, bt: np.ndarray, dev: int = 0,
     436 = 3-buf counted pipeline G16."""
     M, K = a.shape
     N, K2 = bt.shape
-    assert K == K2 and M % 256 == 0 and N % 256 == 0 and K % 256 == 0
+    tile_m = 512 if variant == 456 else 256
+    assert K == K2 and M % tile_m == 0 and N % 256 == 0 and K % 256 == 0
     a4 = _pack_nibbles(to_fp4_e2m1(np.ascontiguousarray(a, dtype=np.float32)), K)
     b4 = _pack_nibbles(to_fp4_e2m1(np.ascontiguousarray(bt, dtype=np.float32)), K)
     out = np.zeros((M, N), dtype=np.float32)
@@ -433,7 +434,7 @@ def gemm_fp8(a: np.ndarray, bt: np.ndarray, dev: int = 0,
     fp32 out); inputs are float32, quantized to e4m3 exactly as consumed."""
     M, K = a.shape
     N, K2 = bt.shape
-    tile_m = 128 if variant == 1 else 256  # all 2xx/3xx variants use 256
+    tile_m = {1: 128, 456: 512}.get(variant, 256)
     assert K == K2 and M % tile_m == 0 and N % 128 == 0 and K % 128 == 0
     a8 = to_fp8_e4m3(np.ascontiguousarray(a, dtype=np.float32))
     b8 = to_fp8_e4m3(np.ascontiguousarray(bt, dtype=np.float32))

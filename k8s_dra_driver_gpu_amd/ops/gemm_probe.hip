@@ -1599,12 +1599,13 @@ gemm_fp4_256x256_g16_kernel(const unsigned char* A, const unsigned char* Bt, flo
 // fp4 3-buffer counted-vmcnt variant: BK=128 (2 MFMA K-depths per step),
 // 3 x 32 KiB LDS, single barrier per step (one tile in flight) — measures
 // whether the 2-buffer dual-barrier serialization is the remaining cost.
-template <int GROUP, int SWZ = 0>
+template <int GROUP, int SWZ = 0, int TBM = 256, int TBN = 256,
+          int TWR = 4, int TWC = 2>
 __device__ __forceinline__ void gemm_fp4_3buf_body(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
-    constexpr int TBM = 256, TBN = 256, BK = 128, NT = 512;
-    constexpr int AI = 2, BJ = 4;
+    constexpr int BK = 128, NT = TWR * TWC * WAVE;
+    constexpr int AI = (TBM / TWR) / 32, BJ = (TBN / TWC) / 32;
     constexpr int RB = BK / 2;  // 64 B rows
     __shared__ unsigned char lds[3 * (TBM + TBN) * RB];
     const int HALF = (TBM + TBN) * RB;
@@ -1631,12 +1632,15 @@ __device__ __forceinline__ void gemm_fp4_3buf_body(
     const int tid = threadIdx.x;
     const int lane = tid & (WAVE - 1);
     const int wid = tid / WAVE;
-    const int wr = (wid >> 1) * 64;
-    const int wc = (wid & 1) * 128;
+    const int wr = (wid / TWC) * (TBM / TWR);
+    const int wc = (wid % TWC) * (TBN / TWC);
 
-    constexpr int ACH = TBM * (RB / 16);  // 1024 chunks
-    constexpr int APH = ACH / NT;         // 2
-    constexpr int S = 2 * APH;            // A + B glds per thread per stage
+    constexpr int ACH = TBM * (RB / 16);
+    constexpr int APH = ACH / NT;
+    constexpr int BCH = TBN * (RB / 16);
+    constexpr int BPH = BCH / NT;
+    static_assert(APH * NT == ACH && BPH * NT == BCH, "phase split");
+    constexpr int S = APH + BPH;  // A + B glds per thread per stage
     auto stage = [&](int buf, int k0) {
         unsigned char* la = ldsA(buf);
         unsigned char* lb = ldsB(buf);
@@ -1652,6 +1656,13 @@ __device__ __forceinline__ void gemm_fp4_3buf_body(
                 (const __attribute__((address_space(1))) void*)ga,
                 (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 16),
                 16, 0, 0);
+        }
+#pragma unroll
+        for (int phase = 0; phase < BPH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (RB / 16);
+            int c = chunk % (RB / 16);
+            if (SWZ) c ^= r & (RB / 16 - 1);
             const unsigned char* gb = &Bt[(size_t)(n0 + r) * (K / 2) + kb0 + c * 16];
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) void*)gb,
@@ -1727,6 +1738,14 @@ extern "C" __global__ void __launch_bounds__(512)
 gemm_fp4_3buf_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
                              int M, int N, int K) {
     gemm_fp4_3buf_body<16, 1>(A, Bt, C, M, N, K);
+}
+
+// 512x256 tile: 16 waves (1024 threads), 3 x 48 KiB LDS, halves the
+// B-panel traffic per output element (underfills the chip below 4096^2)
+extern "C" __global__ void __launch_bounds__(1024)
+gemm_fp4_512_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
+                            int M, int N, int K) {
+    gemm_fp4_3buf_body<16, 1, 512, 256, 8, 2>(A, Bt, C, M, N, K);
 }
 
 // fp8 on the 32x32x64 shape (FMT=0), same 3-buffer counted structure as

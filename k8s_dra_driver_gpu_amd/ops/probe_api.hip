@@ -75,6 +75,7 @@ extern "C" __global__ void gemm_fp8k64_3buf_g16_kernel(const unsigned char*, con
 extern "C" __global__ void gemm_fp8_256x256_g16_swz_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp8_256x256_g16_swz_nb_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp4_3buf_g16_swz_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp4_512_g16_swz_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256x256_d2_swz_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256x256_d2_g4_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256x256_d2_g8_kernel(const short*, const short*, float*, int, int, int);
@@ -597,6 +598,7 @@ static inline void (*fp8_kern_for(int v))(const unsigned char*, const unsigned c
         case 416: return gemm_fp4_256x256_g16_kernel;
         case 436: return gemm_fp4_3buf_g16_kernel;
         case 446: return gemm_fp4_3buf_g16_swz_kernel;
+        case 456: return gemm_fp4_512_g16_swz_kernel;
         case 336: return gemm_fp8k64_3buf_g16_kernel;
         case 326: return gemm_fp8_256x256_g16_swz_kernel;
         case 346: return gemm_fp8_256x256_g16_swz_nb_kernel;
@@ -614,7 +616,7 @@ static inline void fp8_geom_for(int v, int* tm, int* tn, int* threads) {
 double fp_gemm_fp8_tflops_ex(int dev, int size, int iters, int variant) {
     CHK(hipSetDevice(dev));
     int M = size, N = size, K = size;
-    const int fp4 = (variant == 4 || variant == 416 || variant == 436 || variant == 446);
+    const int fp4 = (variant == 4 || variant == 416 || variant == 436 || variant == 446 || variant == 456);
     unsigned char *A, *Bt;
     float* C;
     CHK(hipMalloc(&A, (size_t)M * K / (fp4 ? 2 : 1)));
@@ -727,7 +729,7 @@ double fp_mfma_fp8_tflops(int dev, int inner_iters, int launches) {
 int fp_gemm_fp8_host_ex(int dev, const unsigned char* A, const unsigned char* Bt,
                         float* C, int M, int N, int K, int variant) {
     CHKI(hipSetDevice(dev));
-    const int den = (variant == 4 || variant == 416 || variant == 436 || variant == 446) ? 2 : 1;  // fp4: packed bytes
+    const int den = (variant == 4 || variant == 416 || variant == 436 || variant == 446 || variant == 456) ? 2 : 1;  // fp4: packed bytes
     unsigned char *dA, *dB;
     float* dC;
     CHKI(hipMalloc(&dA, (size_t)M * K / den));

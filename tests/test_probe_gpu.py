@@ -402,10 +402,21 @@ class TestFp4Probe:
         bt = rng.standard_normal((N, K)).astype(np.float32)
         ref = (probe.fp4_e2m1_to_f32(probe.to_fp4_e2m1(a)).astype(np.float64)
                @ probe.fp4_e2m1_to_f32(probe.to_fp4_e2m1(bt)).astype(np.float64).T)
-        for v in (4, 416, 436, 446):
+        for v in (4, 416, 436, 446):  # 456 needs M%512; covered below
             d = probe.gemm_fp4(a, bt, variant=v)
             err = np.abs(d - ref).max() / np.abs(ref).max()
             assert err < 1e-6, f"variant {v}: {err}"
+
+    def test_fp4_512_tile_variant(self, probe):
+        rng = np.random.default_rng(13)
+        M, N, K = 1024, 512, 512
+        a = rng.standard_normal((M, K)).astype(np.float32)
+        bt = rng.standard_normal((N, K)).astype(np.float32)
+        ref = (probe.fp4_e2m1_to_f32(probe.to_fp4_e2m1(a)).astype(np.float64)
+               @ probe.fp4_e2m1_to_f32(probe.to_fp4_e2m1(bt)).astype(np.float64).T)
+        d = probe.gemm_fp4(a, bt, variant=456)
+        err = np.abs(d - ref).max() / np.abs(ref).max()
+        assert err < 1e-6, err
 
     def test_fp8_32x32_shape_variant(self, probe):
         """fp8 on the 32x32x64 shape (variant 336): numerically correct but
