@@ -38,6 +38,9 @@ def parse_args(argv=None):
     p.add_argument("--registry-dir",
                    default=env("PLUGINS_REGISTRY_DIR", "/var/lib/kubelet/plugins_registry"))
     p.add_argument("--cdi-root", default=env("CDI_ROOT", "/var/run/cdi"))
+    p.add_argument("--rocm-driver-root", default=env("ROCM_DRIVER_ROOT", "/opt/rocm"),
+                   help="host ROCm install the CDI specs mount into containers "
+                        "(the nvidia-driver-root analog)")
     p.add_argument("--feature-gates", default=env("FEATURE_GATES", ""))
     p.add_argument("--healthcheck-port", type=int, default=int(env("HEALTHCHECK_PORT", "0")))
     p.add_argument("--metrics-port", type=int, default=int(env("METRICS_PORT", "0")))
@@ -80,7 +83,7 @@ def main(argv=None) -> int:
     metrics = DraMetrics()
     state = DeviceState(
         devicelib=devicelib,
-        cdi=CdiHandler(cdi_root=args.cdi_root),
+        cdi=CdiHandler(cdi_root=args.cdi_root, rocm_root=args.rocm_driver_root),
         checkpoints=CheckpointManager(state_dir),
         state_dir=state_dir,
     )
