@@ -8,7 +8,9 @@ export TMPDIR=/tmp
 OUT=gpurun_out/soak
 mkdir -p "$OUT"
 SOAK_S=${SOAK_S:-600}
-timeout $((SOAK_S + 240)) python - <<PYEOF 2>&1 | tail -30 | tee "$OUT/soak.txt"
+# write directly to the file: a piped tee can hang on EOF if a
+# supervised fabricd child inherits the pipe
+timeout $((SOAK_S + 180)) python -u - > "$OUT/soak.txt" 2>&1 <<PYEOF
 import json, os, statistics, tempfile, time, uuid
 from k8s_dra_driver_gpu_amd.bench.localcluster import LocalCluster
 from k8s_dra_driver_gpu_amd.fabric import probe
@@ -17,9 +19,11 @@ from k8s_dra_driver_gpu_amd.daemon.process import default_fabricctl_path
 import subprocess
 
 SOAK_S = int(os.environ.get("SOAK_S", "600"))
+import atexit
 cluster = LocalCluster(real_devices=os.path.exists("/dev/kfd"),
                        num_gpus=2, partitionable=False,
                        work_dir=tempfile.mkdtemp(prefix="soak-")).start()
+atexit.register(cluster.stop)
 cluster.client.create("computedomains", {
     "apiVersion": "resource.amd.com/v1beta1", "kind": "ComputeDomain",
     "metadata": {"name": "soak-cd", "namespace": "default"},
@@ -103,7 +107,7 @@ while time.monotonic() - t_start < SOAK_S:
             minute = int(el // 60)
             print(f"[{el:5.0f}s] cycles={cycles} p50={statistics.median(lats)*1e3:.2f}ms "
                   f"hbm={probe_reads[-1]:.0f}GB/s fabricd={st} errors={errors}")
-        assert st == "READY", f"fabricd degraded: {st}"
+        assert st.startswith("READY"), f"fabricd degraded: {st}"
 cd = cluster.client.get("computedomains", "soak-cd", "default")
 lats.sort()
 print(json.dumps({
