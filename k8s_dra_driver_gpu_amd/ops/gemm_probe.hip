@@ -1740,8 +1740,12 @@ gemm_fp4_3buf_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt, fl
     gemm_fp4_3buf_body<16, 1>(A, Bt, C, M, N, K);
 }
 
-// 512x256 tile: 16 waves (1024 threads), 3 x 48 KiB LDS, halves the
-// B-panel traffic per output element (underfills the chip below 4096^2)
+// 512x256 tile: 16 waves (1024 threads), 3 x 48 KiB LDS. MEASURED
+// CATASTROPHIC (273/169 TF vs 3524/2944 for the 256x256 3-buf champion,
+// numerics exact — gpurun_out/r2s29): the 1024-thread launch bounds force
+// the compiler to fit 128 acc VGPRs + fragments into a half-size budget,
+// spilling to scratch. Kept as a measured-negative data point; the
+// 256x256 tile remains the fp4 shape.
 extern "C" __global__ void __launch_bounds__(1024)
 gemm_fp4_512_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt, float* C,
                             int M, int N, int K) {

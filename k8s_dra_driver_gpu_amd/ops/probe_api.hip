@@ -610,6 +610,7 @@ static inline void fp8_geom_for(int v, int* tm, int* tn, int* threads) {
         *tm = 256; *tn = 128; *threads = 512; return;
     }
     if (v == 3 || v == 316 || v == 326 || v == 336 || v == 346 || v == 4 || v == 416 || v == 436 || v == 446) { *tm = 256; *tn = 256; *threads = 512; return; }
+    if (v == 456) { *tm = 512; *tn = 256; *threads = 1024; return; }
     *tm = 128; *tn = 128; *threads = 256;
 }
 
