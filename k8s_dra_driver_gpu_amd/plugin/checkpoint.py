@@ -102,7 +102,7 @@ class CheckpointData:
     def set_claim(self, uid: str, claim: PreparedClaim) -> None:
         raw = to_dict(claim)
         self.prepared_claims[uid] = raw
-        self._frags()[uid] = _canonical(raw)
+        self._frags()[uid] = f"{json.dumps(uid)}:{_canonical(raw)}"
 
     def remove_claim(self, uid: str) -> None:
         self.prepared_claims.pop(uid, None)
@@ -126,6 +126,11 @@ class CheckpointData:
         return f
 
     def canonical_payload(self) -> str:
+        # Fragments are cached as complete '"<uid>":<claim-json>' strings so
+        # a store with N standing claims is a sorted join of cached strings
+        # (one fresh serialization for the mutated claim only) — the
+        # whole-checkpoint-rewrite cost at large populations is IO+checksum,
+        # not re-serialization (measured: 2000 standing claims).
         frags = self._frags()
         parts = []
         if self.node_boot_id:
@@ -135,11 +140,12 @@ class CheckpointData:
             for uid in sorted(self.prepared_claims):
                 frag = frags.get(uid)
                 if frag is None:
-                    frag = _canonical(to_dict(self.prepared_claims[uid])
+                    body = _canonical(to_dict(self.prepared_claims[uid])
                                       if not isinstance(self.prepared_claims[uid], dict)
                                       else self.prepared_claims[uid])
+                    frag = f"{json.dumps(uid)}:{body}"
                     frags[uid] = frag
-                entries.append(f"{json.dumps(uid)}:{frag}")
+                entries.append(frag)
             parts.append('"preparedClaims":{' + ",".join(entries) + "}")
         return "{" + ",".join(parts) + "}"
 
