@@ -46,6 +46,16 @@ def _load() -> ctypes.CDLL:
     lib.fp_mfma_fp8_tflops.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int]
     lib.fp_mfma_fp4_tflops.restype = ctypes.c_double
     lib.fp_mfma_fp4_tflops.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int]
+    lib.fp_mfma_fp8_scaled_tile_host.restype = ctypes.c_int
+    lib.fp_mfma_fp8_scaled_tile_host.argtypes = [
+        ctypes.c_int,
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_float),
+        ctypes.c_int,
+    ]
     lib.fp_mfma_fp4_tile_gemm_host.restype = ctypes.c_int
     lib.fp_mfma_fp4_tile_gemm_host.argtypes = [
         ctypes.c_int,
@@ -347,6 +357,54 @@ def _pack_nibbles(nib: np.ndarray, axis_len: int) -> np.ndarray:
     """Pack pairs along the last axis (even = low nibble)."""
     flat = nib.reshape(-1, axis_len)
     return (flat[:, 0::2] | (flat[:, 1::2] << 4)).astype(np.uint8)
+
+
+def mx_quantize_fp8(x: np.ndarray, block: int = 32):
+    """MX quantization along the last axis: per-block E8M0 scale (power of
+    two) + e4m3 elements of x/scale. Returns (elems_u8, scales_u8)."""
+    x = np.asarray(x, dtype=np.float32)
+    assert x.shape[-1] % block == 0
+    xb = x.reshape(*x.shape[:-1], -1, block)
+    absmax = np.abs(xb).max(axis=-1, keepdims=True)
+    # scale = 2^e with absmax/2^e <= 448 (e4m3 max); e8m0 byte = e + 127
+    e = np.ceil(np.log2(np.maximum(absmax, 1e-30) / 448.0))
+    e = np.clip(e, -127, 127)
+    scale = np.exp2(e).astype(np.float32)
+    elems = to_fp8_e4m3((xb / scale).reshape(x.shape))
+    scales = (e.astype(np.int32) + 127).astype(np.uint8).reshape(*x.shape[:-1], -1)
+    return elems, scales
+
+
+def mx_dequantize_fp8(elems: np.ndarray, scales: np.ndarray, block: int = 32):
+    v = fp8_e4m3_to_f32(elems)
+    sc = np.exp2(scales.astype(np.float32) - 127.0)
+    vb = v.reshape(*v.shape[:-1], -1, block)
+    return (vb * sc[..., None]).reshape(v.shape)
+
+
+def mfma_fp8_scaled_tile(a: np.ndarray, b: np.ndarray, dev: int = 0) -> np.ndarray:
+    """D[16,16] = a[16,K] @ b[K,16] through the matrix cores with REAL
+    per-32-block MX (E8M0) scales — the HW-fused dequant+matmul path.
+    a is MX-quantized along K; b along its leading (K) axis per column."""
+    K = a.shape[1]
+    assert a.shape == (16, K) and b.shape == (K, 16) and K % 128 == 0
+    a8, sa = mx_quantize_fp8(a)
+    # quantize B per column along K: transpose to [16][K], quantize, back
+    b8_t, sb = mx_quantize_fp8(np.ascontiguousarray(b.T))
+    b8 = np.ascontiguousarray(b8_t.T)
+    out = np.zeros((16, 16), dtype=np.float32)
+    rc = _load().fp_mfma_fp8_scaled_tile_host(
+        dev,
+        np.ascontiguousarray(a8).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(b8).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(sa).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(sb).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+        K,
+    )
+    if rc < 0:
+        raise ProbeError(f"mfma_fp8_scaled_tile failed with hip error {-rc}")
+    return out
 
 
 def mfma_fp4_tile_gemm(a: np.ndarray, b: np.ndarray, dev: int = 0) -> np.ndarray:

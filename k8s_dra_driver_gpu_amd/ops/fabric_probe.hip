@@ -386,6 +386,85 @@ mfma_fp4_tile_gemm_kernel(const unsigned char* __restrict__ A,  // [32][K] e2m1 
     }
 }
 
+// MX-SCALED fp8 tile verify: REAL per-block E8M0 scales (block = 32
+// elements). Scale-operand lane layout was pinned empirically with
+// mfma_scale_probe_kernel (profiles/r2: scale_probe_w2.txt): the scale
+// byte of lane idx+16*g (byte 0, OPSEL=0) covers the two 16-element
+// k-chunks {0,2}/{4,6}/{1,3}/{5,7} (g=0..3) OF THE REGISTER SLOTS, i.e.
+// hardware scale blocks are NOT the contiguous 32 elements a lane holds
+// under the naive slot->k identity. Placing logical chunk CH[s] at slot
+// s with CH = [0,4,1,5,2,6,3,7] (same permutation for A and B, so the
+// dot product is unchanged) makes scale group g cover exactly the
+// contiguous logical block k in [32g, 32g+32): standard MX-32 blocks.
+extern "C" __global__ void __launch_bounds__(WAVE)
+mfma_fp8_scaled_tile_kernel(const unsigned char* __restrict__ A,   // [16][K] e4m3
+                            const unsigned char* __restrict__ B,   // [K][16] e4m3
+                            const unsigned char* __restrict__ SA,  // [16][K/32] e8m0
+                            const unsigned char* __restrict__ SB,  // [16][K/32] e8m0 (per col)
+                            float* __restrict__ D, int K) {
+    int lane = threadIdx.x & (WAVE - 1);
+    int row = lane & 15;       // A row for a-slices; B column for b-slices
+    int kgrp = lane >> 4;      // scale group g: logical MX block [32g, 32g+32)
+    const int CH[8] = {0, 4, 1, 5, 2, 6, 3, 7};
+    f32x4 acc = {};
+    for (int k0 = 0; k0 < K; k0 += 128) {
+        union {
+            i32x8 v;
+            unsigned char b[32];
+        } a, bb;
+        for (int e = 0; e < 32; ++e) {
+            int slot = kgrp * 2 + (e >> 4);
+            int k = k0 + CH[slot] * 16 + (e & 15);
+            a.b[e] = A[row * K + k];
+            bb.b[e] = B[k * 16 + row];
+        }
+        int blk = k0 / 32 + kgrp;
+        int sa = SA[row * (K / 32) + blk];
+        int sb = SB[row * (K / 32) + blk];  // col == row lane mapping
+        acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            a.v, bb.v, acc, 0, 0, 0, sa, 0, sb);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int out_row = kgrp * 4 + r;
+        D[out_row * 16 + row] = acc[r];
+    }
+}
+
+// Scale-operand layout probe: A = all-ones e4m3, B[k][j] = 1 iff
+// k/32 == j%4 (so D[i][j] = 32 * effective_scale(row i, block j%4)).
+// For probe index L (0..63): lane L supplies scale byte 0x80 (2.0), all
+// other lanes 0x7F (1.0) — first 64 probes perturb scale_a, next 64
+// perturb scale_b. Dout[L][16][16] reveals exactly which (row, block)
+// each lane's scale byte controls.
+extern "C" __global__ void __launch_bounds__(WAVE)
+mfma_scale_probe_kernel(float* __restrict__ Dout) {
+    int lane = threadIdx.x & (WAVE - 1);
+    int row = lane & 15;
+    int kgrp = lane >> 4;
+    union { i32x8 v; unsigned char b[32]; } a, bb;
+    for (int e = 0; e < 32; ++e) a.b[e] = 0x38;  // e4m3 1.0
+    for (int e = 0; e < 32; ++e) {
+        int k = kgrp * 32 + e;   // this lane's B rows: col = row
+        // B[k][col] = 2^-(k/16): every chunk contributes a distinguishable
+        // weight to every column, so multi-chunk scale coverage decodes
+        // from the delta (16-chunk ch adds 16 * 2^-ch when doubled)
+        int ch = k / 16;   // e4m3 2^-ch; ch==7 needs the denormal 0x04
+        bb.b[e] = (unsigned char)(ch < 7 ? 0x38 - 8 * ch : 0x04);
+    }
+    // probes 0..127 perturb one scale byte; probe 128 is the baseline
+    for (int L = 0; L < 129; ++L) {
+        int sa = (L < 64 && lane == L) ? 0x80 : 0x7F;
+        int sb = (L >= 64 && L < 128 && lane == (L - 64)) ? 0x80 : 0x7F;
+        f32x4 acc = {};
+        acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            a.v, bb.v, acc, 0, 0, 0, sa, 0, sb);
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            Dout[L * 256 + (kgrp * 4 + r) * 16 + row] = acc[r];
+    }
+}
+
 // Register-resident fp4 issue-rate ubench (4 independent f32x16 accumulators)
 extern "C" __global__ void __launch_bounds__(PROBE_BLOCK)
 mfma_fp4_loop_kernel(const int* __restrict__ seed, float* __restrict__ sink,

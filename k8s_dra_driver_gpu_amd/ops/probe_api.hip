@@ -43,6 +43,8 @@ extern "C" __global__ void mfma_bf16_loop_kernel(const short*, float*, int);
 extern "C" __global__ void mfma_fp8_loop_kernel(const int*, float*, int);
 extern "C" __global__ void mfma_fp4_loop_kernel(const int*, float*, int);
 extern "C" __global__ void mfma_fp4_tile_gemm_kernel(const unsigned char*, const unsigned char*, float*, int);
+extern "C" __global__ void mfma_fp8_scaled_tile_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, float*, int);
+extern "C" __global__ void mfma_scale_probe_kernel(float*);
 extern "C" __global__ void mfma_bf16_tile_gemm_kernel(const short*, const short*, float*, int);
 extern "C" __global__ void mfma_fp8_tile_gemm_kernel(const unsigned char*, const unsigned char*, float*, int, int);
 extern "C" __global__ void p2p_read_kernel(float4v*, const float4v*, long);
@@ -676,6 +678,41 @@ double fp_mfma_fp4_tflops(int dev, int inner_iters, int launches) {
     hipEventDestroy(t0);
     hipEventDestroy(t1);
     return flops / (ms * 1e9);
+}
+
+int fp_mfma_scale_probe_host(int dev, float* D) {
+    CHKI(hipSetDevice(dev));
+    float* dD;
+    CHKI(hipMalloc(&dD, 129 * 256 * sizeof(float)));
+    hipLaunchKernelGGL(mfma_scale_probe_kernel, dim3(1), dim3(64), 0, 0, dD);
+    CHKI(hipGetLastError());
+    CHKI(hipDeviceSynchronize());
+    CHKI(hipMemcpy(D, dD, 129 * 256 * sizeof(float), hipMemcpyDeviceToHost));
+    hipFree(dD);
+    return 0;
+}
+
+int fp_mfma_fp8_scaled_tile_host(int dev, const unsigned char* A, const unsigned char* B,
+                                 const unsigned char* SA, const unsigned char* SB,
+                                 float* D, int K) {
+    CHKI(hipSetDevice(dev));
+    unsigned char *dA, *dB, *dSA, *dSB;
+    float* dD;
+    CHKI(hipMalloc(&dA, (size_t)16 * K));
+    CHKI(hipMalloc(&dB, (size_t)K * 16));
+    CHKI(hipMalloc(&dSA, (size_t)16 * K / 32));
+    CHKI(hipMalloc(&dSB, (size_t)16 * K / 32));
+    CHKI(hipMalloc(&dD, 16 * 16 * sizeof(float)));
+    CHKI(hipMemcpy(dA, A, (size_t)16 * K, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dB, B, (size_t)K * 16, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dSA, SA, (size_t)16 * K / 32, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dSB, SB, (size_t)16 * K / 32, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(mfma_fp8_scaled_tile_kernel, dim3(1), dim3(64), 0, 0, dA, dB, dSA, dSB, dD, K);
+    CHKI(hipGetLastError());
+    CHKI(hipDeviceSynchronize());
+    CHKI(hipMemcpy(D, dD, 16 * 16 * sizeof(float), hipMemcpyDeviceToHost));
+    hipFree(dA); hipFree(dB); hipFree(dSA); hipFree(dSB); hipFree(dD);
+    return 0;
 }
 
 int fp_mfma_fp4_tile_gemm_host(int dev, const unsigned char* A, const unsigned char* B,
