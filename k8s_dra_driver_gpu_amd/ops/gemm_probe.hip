@@ -1154,6 +1154,7 @@ gemm_bf16_256x256_d2_g8_kernel(const short* A, const short* Bt, float* C, int M,
 // ---------------------------------------------------------------------------
 
 typedef int i32x8 __attribute__((ext_vector_type(8)));
+typedef int i32x4g __attribute__((ext_vector_type(4)));
 
 template <int TBM, int TBN, int TWR, int TWC, int GROUP = 1>
 __device__ __forceinline__ void gemm_fp8_body(
@@ -1279,6 +1280,383 @@ __device__ __forceinline__ void gemm_fp8_body(
                 int col = n0 + wc + j * 16 + (lane & 15);
                 if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
             }
+}
+
+// MX-scaled fp8 GEMM: real per-block E8M0 scales through the mfma scale
+// operands. Same 3-buffer global_load_lds pipeline as gemm_fp8_body; two
+// differences, both forced by the hardware scale-lane layout pinned in
+// profiles/r2/scale_probe_w2.txt:
+//  * a lane in scale group g (= lane>>4) must hold logical k-chunks
+//    {g, g+4} in its register slots (CH = [0,4,1,5,2,6,3,7]), so the
+//    fragment is two 16-byte LDS reads instead of one 32-byte read;
+//  * each mfma takes sa = SA[row][blk], sb = SB[col][blk] at byte 0
+//    (OPSEL 0), blk = k0/32 + g. Scale bytes are read straight from
+//    global: per step a lane loads AI+BJ bytes against 16 mfmas, and the
+//    per-tile scale slice (TBM+TBN bytes per k-step) lives in L2.
+template <int TBM, int TBN, int TWR, int TWC, int GROUP = 1, int SWZ = 0, int LOADS = 1>
+__device__ __forceinline__ void gemm_fp8_scaled_body(
+    const unsigned char* __restrict__ A,   // [M][K] row-major e4m3
+    const unsigned char* __restrict__ Bt,  // [N][K] row-major e4m3
+    const unsigned char* __restrict__ SA,  // [M][K/32] e8m0
+    const unsigned char* __restrict__ SBt, // [N][K/32] e8m0
+    float* __restrict__ C,                 // [M][N] row-major f32
+    int M, int N, int K) {
+    constexpr int BK = 128;
+    constexpr int NT = TWR * TWC * WAVE;
+    constexpr int WM = TBM / TWR;
+    constexpr int WN = TBN / TWC;
+    constexpr int AI = WM / 16;
+    constexpr int BJ = WN / 16;
+    // LOADS==2: per-buffer scale panel ((TBM+TBN)*4 B) staged next to data
+    constexpr int SCB = (LOADS == 2) ? (TBM + TBN) * 4 : 0;
+    __shared__ unsigned char lds[3 * (TBM * BK + TBN * BK + SCB)];
+    const int HALF = TBM * BK + TBN * BK + SCB;
+    auto ldsA = [&](int buf) -> unsigned char* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> unsigned char* { return lds + buf * HALF + TBM * BK; };
+    auto ldsS = [&](int buf) -> unsigned char* { return lds + buf * HALF + TBM * BK + TBN * BK; };
+
+    const int tiles_n = (N + TBN - 1) / TBN;
+    int tile_m, tile_n;
+    if (GROUP > 1) {
+        const int tiles_m = (M + TBM - 1) / TBM;
+        const int per_group = GROUP * tiles_n;
+        const int gid = blockIdx.x / per_group;
+        const int first_m = gid * GROUP;
+        const int gsz = min(GROUP, tiles_m - first_m);
+        tile_m = first_m + (blockIdx.x % per_group) % gsz;
+        tile_n = (blockIdx.x % per_group) / gsz;
+    } else {
+        tile_m = blockIdx.x / tiles_n;
+        tile_n = blockIdx.x % tiles_n;
+    }
+    const int m0 = tile_m * TBM;
+    const int n0 = tile_n * TBN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid / TWC) * WM;
+    const int wc = (wid % TWC) * WN;
+    const int kblocks_ = K / 32;
+
+    constexpr int ACH = TBM * (BK / 16);
+    constexpr int APH = ACH / NT;
+    constexpr int BCH = TBN * (BK / 16);
+    constexpr int BPH = BCH / NT;
+    static_assert(APH * NT == ACH && BPH * NT == BCH, "phase split");
+    constexpr int S = APH + BPH;
+    auto stage = [&](int buf, int k0) {
+        unsigned char* la = ldsA(buf);
+        unsigned char* lb = ldsB(buf);
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 16);
+            int c = chunk % (BK / 16);
+            if (SWZ) c ^= r & 7;  // chunk-granular XOR: spreads LDS banks
+            const unsigned char* ga = &A[(size_t)(m0 + r) * K + k0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+#pragma unroll
+        for (int phase = 0; phase < BPH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 16);
+            int c = chunk % (BK / 16);
+            if (SWZ) c ^= r & 7;
+            const unsigned char* gb = &Bt[(size_t)(n0 + r) * K + k0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+        if (LOADS == 2 && tid < TBM + TBN) {
+            // this k-step's 4 scale bytes for one A row / B column. A
+            // masked instruction still bumps vmcnt for waves with any
+            // active lane, which only makes the vmcnt(S) waits stricter.
+            const unsigned char* gs = tid < TBM
+                ? &SA[(size_t)(m0 + tid) * kblocks_ + k0 / 32]
+                : &SBt[(size_t)(n0 + (tid - TBM)) * kblocks_ + k0 / 32];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gs,
+                (__attribute__((address_space(3))) void*)(ldsS(buf) + tid * 4),
+                4, 0, 0);
+        }
+    };
+
+    f32x4 acc[AI][BJ] = {};
+    const int g = lane >> 4;  // scale group: logical MX block k0/32 + g
+    const int kblocks = K / 32;
+    const int steps = K / BK;
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s % 3;
+        if (s + 1 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 2 < steps) stage((s + 2) % 3, (s + 2) * BK);
+
+        const unsigned char* la = ldsA(buf);
+        const unsigned char* lb = ldsB(buf);
+        const int blk = s * 4 + g;
+        union frag { i32x8 v; i32x4g h[2]; };
+        frag af[AI], bf[BJ];
+        int sa[AI], sb[BJ];
+#pragma unroll
+        for (int i = 0; i < AI; ++i) {
+            const int ar = wr + i * 16 + (lane & 15);
+            const int x = SWZ ? (ar & 7) : 0;
+            af[i].h[0] = *(const i32x4g*)&la[ar * BK + (g ^ x) * 16];
+            af[i].h[1] = *(const i32x4g*)&la[ar * BK + ((g + 4) ^ x) * 16];
+            sa[i] = LOADS == 2 ? ldsS(buf)[ar * 4 + g]
+                  : LOADS       ? SA[(size_t)(m0 + ar) * kblocks + blk] : 0x7F;
+        }
+#pragma unroll
+        for (int j = 0; j < BJ; ++j) {
+            const int bc = wc + j * 16 + (lane & 15);
+            const int x = SWZ ? (bc & 7) : 0;
+            bf[j].h[0] = *(const i32x4g*)&lb[bc * BK + (g ^ x) * 16];
+            bf[j].h[1] = *(const i32x4g*)&lb[bc * BK + ((g + 4) ^ x) * 16];
+            sb[j] = LOADS == 2 ? ldsS(buf)[TBM * 4 + bc * 4 + g]
+                  : LOADS       ? SBt[(size_t)(n0 + bc) * kblocks + blk] : 0x7F;
+        }
+#pragma unroll
+        for (int i = 0; i < AI; ++i)
+#pragma unroll
+            for (int j = 0; j < BJ; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                    af[i].v, bf[j].v, acc[i][j], 0, 0, 0, sa[i], 0, sb[j]);
+    }
+
+#pragma unroll
+    for (int i = 0; i < AI; ++i)
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr + i * 16 + (lane >> 4) * 4 + r;
+                int col = n0 + wc + j * 16 + (lane & 15);
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+gemm_fp8_scaled_128_kernel(const unsigned char* A, const unsigned char* Bt,
+                           const unsigned char* SA, const unsigned char* SBt,
+                           float* C, int M, int N, int K) {
+    gemm_fp8_scaled_body<128, 128, 2, 2>(A, Bt, SA, SBt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_scaled_256_g16_kernel(const unsigned char* A, const unsigned char* Bt,
+                               const unsigned char* SA, const unsigned char* SBt,
+                               float* C, int M, int N, int K) {
+    gemm_fp8_scaled_body<256, 128, 4, 2, 16>(A, Bt, SA, SBt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_scaled_256_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt,
+                                   const unsigned char* SA, const unsigned char* SBt,
+                                   float* C, int M, int N, int K) {
+    gemm_fp8_scaled_body<256, 128, 4, 2, 16, 1>(A, Bt, SA, SBt, C, M, N, K);
+}
+
+// diagnostic: identical split-chunk data path, scales hardcoded to 1.0 —
+// isolates the cost of the per-step scale loads from the fragment split
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_scaled_noload_kernel(const unsigned char* A, const unsigned char* Bt,
+                              const unsigned char* SA, const unsigned char* SBt,
+                              float* C, int M, int N, int K) {
+    gemm_fp8_scaled_body<256, 128, 4, 2, 16, 1, 0>(A, Bt, SA, SBt, C, M, N, K);
+}
+
+// Production MX-scaled body: K-loop grouped by 4 steps so each lane loads
+// its scale bytes as ONE 16-byte vector per row per 4 steps (16 blocks =
+// 512 k) instead of a byte per step — measured 3x: the per-step scalar
+// byte loads were the whole gap to the unscaled champion (529 diagnostic:
+// 2205 TF no-loads vs 747 byte-loads). Requires K % 512 == 0.
+template <int TBM, int TBN, int TWR, int TWC, int GROUP, int SWZ>
+__device__ __forceinline__ void gemm_fp8_scaled4_body(
+    const unsigned char* __restrict__ A,
+    const unsigned char* __restrict__ Bt,
+    const unsigned char* __restrict__ SA,
+    const unsigned char* __restrict__ SBt,
+    float* __restrict__ C, int M, int N, int K) {
+    constexpr int BK = 128;
+    constexpr int NT = TWR * TWC * WAVE;
+    constexpr int WM = TBM / TWR;
+    constexpr int WN = TBN / TWC;
+    constexpr int AI = WM / 16;
+    constexpr int BJ = WN / 16;
+    __shared__ unsigned char lds[3 * (TBM * BK + TBN * BK)];
+    const int HALF = TBM * BK + TBN * BK;
+    auto ldsA = [&](int buf) -> unsigned char* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> unsigned char* { return lds + buf * HALF + TBM * BK; };
+
+    const int tiles_n = (N + TBN - 1) / TBN;
+    int tile_m, tile_n;
+    if (GROUP > 1) {
+        const int tiles_m = (M + TBM - 1) / TBM;
+        const int per_group = GROUP * tiles_n;
+        const int gid = blockIdx.x / per_group;
+        const int first_m = gid * GROUP;
+        const int gsz = min(GROUP, tiles_m - first_m);
+        tile_m = first_m + (blockIdx.x % per_group) % gsz;
+        tile_n = (blockIdx.x % per_group) / gsz;
+    } else {
+        tile_m = blockIdx.x / tiles_n;
+        tile_n = blockIdx.x % tiles_n;
+    }
+    const int m0 = tile_m * TBM;
+    const int n0 = tile_n * TBN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid / TWC) * WM;
+    const int wc = (wid % TWC) * WN;
+
+    constexpr int ACH = TBM * (BK / 16);
+    constexpr int APH = ACH / NT;
+    constexpr int BCH = TBN * (BK / 16);
+    constexpr int BPH = BCH / NT;
+    static_assert(APH * NT == ACH && BPH * NT == BCH, "phase split");
+    constexpr int S = APH + BPH;
+    auto stage = [&](int buf, int k0) {
+        unsigned char* la = ldsA(buf);
+        unsigned char* lb = ldsB(buf);
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 16);
+            int c = chunk % (BK / 16);
+            if (SWZ) c ^= r & 7;
+            const unsigned char* ga = &A[(size_t)(m0 + r) * K + k0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+#pragma unroll
+        for (int phase = 0; phase < BPH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (BK / 16);
+            int c = chunk % (BK / 16);
+            if (SWZ) c ^= r & 7;
+            const unsigned char* gb = &Bt[(size_t)(n0 + r) * K + k0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+    };
+
+    f32x4 acc[AI][BJ] = {};
+    const int g = lane >> 4;
+    const int kblocks = K / 32;
+    const int steps = K / BK;   // must be a multiple of 4
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    for (int s4 = 0; s4 < steps; s4 += 4) {
+        // 16 scale blocks (= these 4 k-steps) per affected row, one 16B load
+        union sv { i32x4g v; unsigned char b[16]; };
+        sv sad[AI], sbd[BJ];
+#pragma unroll
+        for (int i = 0; i < AI; ++i) {
+            const int ar = wr + i * 16 + (lane & 15);
+            sad[i].v = *(const i32x4g*)&SA[(size_t)(m0 + ar) * kblocks + s4 * 4];
+        }
+#pragma unroll
+        for (int j = 0; j < BJ; ++j) {
+            const int bc = wc + j * 16 + (lane & 15);
+            sbd[j].v = *(const i32x4g*)&SBt[(size_t)(n0 + bc) * kblocks + s4 * 4];
+        }
+#pragma unroll
+        for (int ss = 0; ss < 4; ++ss) {
+            const int s = s4 + ss;
+            const int buf = s % 3;
+            // strict count: the 8 group-start scale loads sit between
+            // stage(s4+1) and stage(s4+2) in issue order, so vmcnt(S)
+            // remains sufficient at every inner step (at ss=0/1 it also
+            // drains the scale loads — one L2 latency per 4 steps)
+            if (s + 1 < steps)
+                asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+            if (s + 2 < steps) stage((s + 2) % 3, (s + 2) * BK);
+
+            const unsigned char* la = ldsA(buf);
+            const unsigned char* lb = ldsB(buf);
+            union frag { i32x8 v; i32x4g h[2]; };
+            frag af[AI], bf[BJ];
+            int sa[AI], sb[BJ];
+            (void)0;
+#pragma unroll
+            for (int i = 0; i < AI; ++i) {
+                const int ar = wr + i * 16 + (lane & 15);
+                const int x = SWZ ? (ar & 7) : 0;
+                af[i].h[0] = *(const i32x4g*)&la[ar * BK + (g ^ x) * 16];
+                af[i].h[1] = *(const i32x4g*)&la[ar * BK + ((g + 4) ^ x) * 16];
+            }
+#pragma unroll
+            for (int j = 0; j < BJ; ++j) {
+                const int bc = wc + j * 16 + (lane & 15);
+                const int x = SWZ ? (bc & 7) : 0;
+                bf[j].h[0] = *(const i32x4g*)&lb[bc * BK + (g ^ x) * 16];
+                bf[j].h[1] = *(const i32x4g*)&lb[bc * BK + ((g + 4) ^ x) * 16];
+            }
+            // byte ss*4 + g of the 16B scale vector = block s*4+g: dword ss
+            // shifted by 8*g (g is per-lane, so a variable shift not a
+            // dynamic byte index — no scratch)
+#pragma unroll
+            for (int i = 0; i < AI; ++i)
+                sa[i] = (((const unsigned*)&sad[i].v)[ss] >> (8 * g)) & 0xFF;
+#pragma unroll
+            for (int j = 0; j < BJ; ++j)
+                sb[j] = (((const unsigned*)&sbd[j].v)[ss] >> (8 * g)) & 0xFF;
+#pragma unroll
+            for (int i = 0; i < AI; ++i)
+#pragma unroll
+                for (int j = 0; j < BJ; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                        af[i].v, bf[j].v, acc[i][j], 0, 0, 0, sa[i], 0, sb[j]);
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < AI; ++i)
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr + i * 16 + (lane >> 4) * 4 + r;
+                int col = n0 + wc + j * 16 + (lane & 15);
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_scaled_slds_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt,
+                                    const unsigned char* SA, const unsigned char* SBt,
+                                    float* C, int M, int N, int K) {
+    gemm_fp8_scaled_body<256, 128, 4, 2, 16, 1, 2>(A, Bt, SA, SBt, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp8_scaled4_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt,
+                                const unsigned char* SA, const unsigned char* SBt,
+                                float* C, int M, int N, int K) {
+    gemm_fp8_scaled4_body<256, 128, 4, 2, 16, 1>(A, Bt, SA, SBt, C, M, N, K);
 }
 
 // 128x128: 3 x 32 KiB LDS (2 WG/CU), 4 waves of 64x64

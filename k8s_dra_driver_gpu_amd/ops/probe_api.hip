@@ -76,6 +76,12 @@ extern "C" __global__ void gemm_fp4_3buf_g16_kernel(const unsigned char*, const 
 extern "C" __global__ void gemm_fp8k64_3buf_g16_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp8_256x256_g16_swz_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp8_256x256_g16_swz_nb_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_scaled_128_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_scaled_256_g16_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_scaled_256_g16_swz_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_scaled_noload_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_scaled4_g16_swz_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, float*, int, int, int);
+extern "C" __global__ void gemm_fp8_scaled_slds_g16_swz_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp4_3buf_g16_swz_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_fp4_512_g16_swz_kernel(const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256x256_d2_swz_kernel(const short*, const short*, float*, int, int, int);
@@ -713,6 +719,84 @@ int fp_mfma_fp8_scaled_tile_host(int dev, const unsigned char* A, const unsigned
     CHKI(hipMemcpy(D, dD, 16 * 16 * sizeof(float), hipMemcpyDeviceToHost));
     hipFree(dA); hipFree(dB); hipFree(dSA); hipFree(dSB); hipFree(dD);
     return 0;
+}
+
+// MX-scaled fp8 GEMM: variant 5 = 128x128/256t, 52 = 256x128 G16/512t
+static inline void (*fp8s_kern_for(int v))(const unsigned char*, const unsigned char*,
+                                           const unsigned char*, const unsigned char*,
+                                           float*, int, int, int) {
+    if (v == 526) return gemm_fp8_scaled_256_g16_swz_kernel;
+    if (v == 529) return gemm_fp8_scaled_noload_kernel;
+    if (v == 546) return gemm_fp8_scaled4_g16_swz_kernel;
+    if (v == 556) return gemm_fp8_scaled_slds_g16_swz_kernel;
+    return v == 52 ? gemm_fp8_scaled_256_g16_kernel : gemm_fp8_scaled_128_kernel;
+}
+static inline void fp8s_geom_for(int v, int* tm, int* tn, int* threads) {
+    if (v == 52 || v == 526 || v == 529 || v == 546 || v == 556) { *tm = 256; *tn = 128; *threads = 512; return; }
+    *tm = 128; *tn = 128; *threads = 256;
+}
+
+int fp_gemm_fp8_scaled_host(int dev, const unsigned char* A, const unsigned char* Bt,
+                            const unsigned char* SA, const unsigned char* SBt,
+                            float* C, int M, int N, int K, int variant) {
+    CHKI(hipSetDevice(dev));
+    unsigned char *dA, *dB, *dSA, *dSB;
+    float* dC;
+    CHKI(hipMalloc(&dA, (size_t)M * K));
+    CHKI(hipMalloc(&dB, (size_t)N * K));
+    CHKI(hipMalloc(&dSA, (size_t)M * K / 32));
+    CHKI(hipMalloc(&dSB, (size_t)N * K / 32));
+    CHKI(hipMalloc(&dC, (size_t)M * N * sizeof(float)));
+    CHKI(hipMemcpy(dA, A, (size_t)M * K, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dB, Bt, (size_t)N * K, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dSA, SA, (size_t)M * K / 32, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dSB, SBt, (size_t)N * K / 32, hipMemcpyHostToDevice));
+    int tm, tn, threads;
+    fp8s_geom_for(variant, &tm, &tn, &threads);
+    dim3 grid(((M + tm - 1) / tm) * ((N + tn - 1) / tn));
+    hipLaunchKernelGGL(fp8s_kern_for(variant), grid, dim3(threads), 0, 0,
+                       dA, dB, dSA, dSB, dC, M, N, K);
+    CHKI(hipGetLastError());
+    CHKI(hipDeviceSynchronize());
+    CHKI(hipMemcpy(C, dC, (size_t)M * N * sizeof(float), hipMemcpyDeviceToHost));
+    hipFree(dA); hipFree(dB); hipFree(dSA); hipFree(dSB); hipFree(dC);
+    return 0;
+}
+
+double fp_gemm_fp8_scaled_tflops(int dev, int size, int iters, int variant) {
+    CHK(hipSetDevice(dev));
+    int M = size, N = size, K = size;
+    unsigned char *A, *Bt, *SA, *SBt;
+    float* C;
+    CHK(hipMalloc(&A, (size_t)M * K));
+    CHK(hipMalloc(&Bt, (size_t)N * K));
+    CHK(hipMalloc(&SA, (size_t)M * K / 32));
+    CHK(hipMalloc(&SBt, (size_t)N * K / 32));
+    CHK(hipMalloc(&C, (size_t)M * N * sizeof(float)));
+    CHK(hipMemset(A, 0x38, (size_t)M * K));
+    CHK(hipMemset(Bt, 0x30, (size_t)N * K));
+    CHK(hipMemset(SA, 0x7F, (size_t)M * K / 32));   // e8m0 1.0
+    CHK(hipMemset(SBt, 0x7F, (size_t)N * K / 32));
+    int tm, tn, threads;
+    fp8s_geom_for(variant, &tm, &tn, &threads);
+    dim3 grid((M / tm) * (N / tn));
+    auto kern = fp8s_kern_for(variant);
+    hipEvent_t t0, t1;
+    CHK(hipEventCreate(&t0));
+    CHK(hipEventCreate(&t1));
+    hipLaunchKernelGGL(kern, grid, dim3(threads), 0, 0, A, Bt, SA, SBt, C, M, N, K);
+    CHK(hipGetLastError());
+    CHK(hipDeviceSynchronize());
+    CHK(hipEventRecord(t0));
+    for (int i = 0; i < iters; ++i)
+        hipLaunchKernelGGL(kern, grid, dim3(threads), 0, 0, A, Bt, SA, SBt, C, M, N, K);
+    CHK(hipEventRecord(t1));
+    CHK(hipEventSynchronize(t1));
+    double ms = time_kernel_ms(t0, t1);
+    hipFree(A); hipFree(Bt); hipFree(SA); hipFree(SBt); hipFree(C);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    return 2.0 * M * (double)N * K * iters / (ms * 1e9);
 }
 
 int fp_mfma_fp4_tile_gemm_host(int dev, const unsigned char* A, const unsigned char* B,

@@ -370,6 +370,22 @@ class TestFp8Probe:
         err = np.abs(d - ref).max() / np.abs(ref).max()
         assert err < 1e-4, err
 
+    def test_mx_scaled_gemm_numerics(self, probe):
+        # full scaled GEMM (LDS-staged scale panels, variant 556 default)
+        # vs the exact dequantized MX reference
+        rng = np.random.default_rng(34)
+        M = N = K = 512
+        a = (rng.standard_normal((M, K))
+             * np.exp2(rng.integers(-6, 7, (M, K)))).astype(np.float32)
+        bt = (rng.standard_normal((N, K))
+              * np.exp2(rng.integers(-6, 7, (N, K)))).astype(np.float32)
+        for v in (52, 526, 556):
+            c, a8, sa, b8t, sbt = probe.gemm_fp8_scaled(a, bt, variant=v)
+            ref = (probe.mx_dequantize_fp8(a8, sa).astype(np.float64)
+                   @ probe.mx_dequantize_fp8(b8t, sbt).astype(np.float64).T)
+            err = np.abs(c - ref).max() / np.abs(ref).max()
+            assert err < 1e-4, f"variant {v}: {err}"
+
     def test_fp8_codec_round_trip(self, probe):
         # CPU-only property of the host codec, kept here with the fp8 suite
         rng = np.random.default_rng(0)
