@@ -56,6 +56,28 @@ def _load() -> ctypes.CDLL:
         ctypes.POINTER(ctypes.c_float),
         ctypes.c_int,
     ]
+    lib.fp_mfma_fp4_scaled_tile_host.restype = ctypes.c_int
+    lib.fp_mfma_fp4_scaled_tile_host.argtypes = [
+        ctypes.c_int,
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_float),
+        ctypes.c_int,
+    ]
+    lib.fp_gemm_fp4_scaled_host.restype = ctypes.c_int
+    lib.fp_gemm_fp4_scaled_host.argtypes = [
+        ctypes.c_int,
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_uint8),
+        ctypes.POINTER(ctypes.c_float),
+        ctypes.c_int, ctypes.c_int, ctypes.c_int,
+    ]
+    lib.fp_gemm_fp4_scaled_tflops.restype = ctypes.c_double
+    lib.fp_gemm_fp4_scaled_tflops.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int]
     lib.fp_mfma_fp4_tile_gemm_host.restype = ctypes.c_int
     lib.fp_mfma_fp4_tile_gemm_host.argtypes = [
         ctypes.c_int,
@@ -394,6 +416,86 @@ def mx_dequantize_fp8(elems: np.ndarray, scales: np.ndarray, block: int = 32):
     sc = np.exp2(scales.astype(np.float32) - 127.0)
     vb = v.reshape(*v.shape[:-1], -1, block)
     return (vb * sc[..., None]).reshape(v.shape)
+
+
+def mx_quantize_fp4(x: np.ndarray, block: int = 32):
+    """MX quantization along the last axis for e2m1: per-block E8M0 scale
+    with absmax/scale <= 6.0 (e2m1 max). Returns (nibbles_u8, scales_u8)."""
+    x = np.asarray(x, dtype=np.float32)
+    assert x.shape[-1] % block == 0
+    xb = x.reshape(*x.shape[:-1], -1, block)
+    absmax = np.abs(xb).max(axis=-1, keepdims=True)
+    e = np.ceil(np.log2(np.maximum(absmax, 1e-30) / 6.0))
+    e = np.clip(e, -127, 127)
+    scale = np.exp2(e).astype(np.float32)
+    nib = to_fp4_e2m1((xb / scale).reshape(x.shape))
+    scales = (e.astype(np.int32) + 127).astype(np.uint8).reshape(*x.shape[:-1], -1)
+    return nib, scales
+
+
+def mx_dequantize_fp4(nib: np.ndarray, scales: np.ndarray, block: int = 32):
+    v = fp4_e2m1_to_f32(nib)
+    sc = np.exp2(scales.astype(np.float32) - 127.0)
+    vb = v.reshape(*v.shape[:-1], -1, block)
+    return (vb * sc[..., None]).reshape(v.shape)
+
+
+def mfma_fp4_scaled_tile(a: np.ndarray, b: np.ndarray, dev: int = 0) -> np.ndarray:
+    """D[32,32] = a[32,K] @ b[K,32] with REAL per-32-block MX scales on the
+    fp4 matrix cores (naive scale layout, pinned by the fp4 scale probe)."""
+    K = a.shape[1]
+    assert a.shape == (32, K) and b.shape == (K, 32) and K % 64 == 0
+    a4, sa = mx_quantize_fp4(a)
+    b4t, sb = mx_quantize_fp4(np.ascontiguousarray(b.T))
+    b4 = np.ascontiguousarray(b4t.T)
+    a_packed = _pack_nibbles(a4, K)
+    b_pairs = b4.reshape(K // 2, 2, 32)
+    b_packed = (b_pairs[:, 0, :] | (b_pairs[:, 1, :] << 4)).astype(np.uint8)
+    out = np.zeros((32, 32), dtype=np.float32)
+    rc = _load().fp_mfma_fp4_scaled_tile_host(
+        dev,
+        np.ascontiguousarray(a_packed).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(b_packed).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(sa).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(sb).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+        K,
+    )
+    if rc < 0:
+        raise ProbeError(f"mfma_fp4_scaled_tile failed with hip error {-rc}")
+    return out
+
+
+def gemm_fp4_scaled(a: np.ndarray, bt: np.ndarray, dev: int = 0):
+    """C[M,N] = MX-dequant(a) @ MX-dequant(bt)^T, per-block E8M0 scales
+    applied by the fp4 matrix cores (256x256 G16 XOR-swizzled tiles with
+    LDS-staged scale panels). Returns (C, a4, sa, b4t, sbt)."""
+    M, K = a.shape
+    N, K2 = bt.shape
+    assert K == K2 and M % 256 == 0 and N % 256 == 0 and K % 128 == 0
+    a4, sa = mx_quantize_fp4(np.ascontiguousarray(a, dtype=np.float32))
+    b4t, sbt = mx_quantize_fp4(np.ascontiguousarray(bt, dtype=np.float32))
+    a_packed = _pack_nibbles(a4, K)
+    b_packed = _pack_nibbles(b4t, K)
+    out = np.zeros((M, N), dtype=np.float32)
+    rc = _load().fp_gemm_fp4_scaled_host(
+        dev,
+        np.ascontiguousarray(a_packed).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(b_packed).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(sa).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        np.ascontiguousarray(sbt).ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+        M, N, K,
+    )
+    if rc < 0:
+        raise ProbeError(f"gemm_fp4_scaled failed with hip error {-rc}")
+    return out, a4, sa, b4t, sbt
+
+
+def gemm_fp4_scaled_tflops(dev: int = 0, size: int = 4096, iters: int = 10) -> float:
+    """MX-scaled fp4 GEMM throughput (real per-block scales in the loop)."""
+    return _check(_load().fp_gemm_fp4_scaled_tflops(dev, size, iters),
+                  "gemm_fp4_scaled_tflops")
 
 
 def mfma_fp8_scaled_tile(a: np.ndarray, b: np.ndarray, dev: int = 0) -> np.ndarray:

@@ -465,6 +465,81 @@ mfma_scale_probe_kernel(float* __restrict__ Dout) {
     }
 }
 
+// fp4 tile GEMM with REAL per-block MX scales: the fp4 scale layout is
+// naive (lane idx+32*kgrp scales its own k in [32*kgrp, 32*kgrp+32) —
+// probe profiles/r2/fp4_scale_probe.txt), so sa = SA[row][k0/32+kgrp].
+extern "C" __global__ void __launch_bounds__(WAVE)
+mfma_fp4_scaled_tile_kernel(const unsigned char* __restrict__ A,   // [32][K/2] packed
+                            const unsigned char* __restrict__ B,   // [K/2][32] packed along k
+                            const unsigned char* __restrict__ SA,  // [32][K/32] e8m0
+                            const unsigned char* __restrict__ SB,  // [32][K/32] e8m0 per col
+                            float* __restrict__ D, int K) {
+    int lane = threadIdx.x & (WAVE - 1);
+    int col = lane & 31;
+    int kgrp = lane >> 5;
+    f32x16 acc = {};
+    for (int k0 = 0; k0 < K; k0 += 64) {
+        union { i32x8 v; unsigned char b[32]; } a, bb;
+        for (int e = 0; e < 32; ++e) a.b[e] = 0, bb.b[e] = 0;
+        for (int j = 0; j < 32; ++j) {
+            int k = k0 + kgrp * 32 + j;
+            unsigned char av = A[(size_t)col * (K / 2) + k / 2];
+            unsigned char an = (k & 1) ? (av >> 4) : (av & 0xF);
+            unsigned char bv = B[(size_t)(k / 2) * 32 + col];
+            unsigned char bn = (k & 1) ? (bv >> 4) : (bv & 0xF);
+            if (j & 1) {
+                a.b[j / 2] |= an << 4;
+                bb.b[j / 2] |= bn << 4;
+            } else {
+                a.b[j / 2] |= an;
+                bb.b[j / 2] |= bn;
+            }
+        }
+        int blk = k0 / 32 + kgrp;
+        int sa = SA[(size_t)col * (K / 32) + blk];
+        int sb = SB[(size_t)col * (K / 32) + blk];
+        acc = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            a.v, bb.v, acc, 4, 4, 0, sa, 0, sb);
+    }
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+        int row = (r & 3) + 8 * (r >> 2) + 4 * kgrp;
+        D[row * 32 + col] = acc[r];
+    }
+}
+
+// Scale-operand layout probe for the fp4 32x32x64 shape: A = all 1.0
+// e2m1; B[k][col] = w(k/16) with w = {4,2,1,0.5} so every 16-chunk
+// contributes a distinguishable weight (delta of a doubled chunk ch is
+// 16*w(ch) in {64,32,16,8}). Probes 0..63 double SA lane L's byte 0,
+// 64..127 double SB lane L-64's, probe 128 is the baseline.
+extern "C" __global__ void __launch_bounds__(WAVE)
+mfma_fp4_scale_probe_kernel(float* __restrict__ Dout) {
+    int lane = threadIdx.x & (WAVE - 1);
+    int kgrp = lane >> 5;
+    union { i32x8 v; unsigned char b[32]; } a, bb;
+    const unsigned char W[4] = {0x6, 0x4, 0x2, 0x1};  // e2m1 4,2,1,0.5
+    for (int e = 0; e < 32; ++e) a.b[e] = 0, bb.b[e] = 0;
+    for (int j = 0; j < 32; j += 2) {
+        int k = kgrp * 32 + j;
+        a.b[j / 2] = 0x22;  // two 1.0 nibbles
+        bb.b[j / 2] = (unsigned char)(W[k / 16] | (W[(k + 1) / 16] << 4));
+    }
+    int col = lane & 31;
+    for (int L = 0; L < 129; ++L) {
+        int sa = (L < 64 && lane == L) ? 0x80 : 0x7F;
+        int sb = (L >= 64 && L < 128 && lane == (L - 64)) ? 0x80 : 0x7F;
+        f32x16 acc = {};
+        acc = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            a.v, bb.v, acc, 4, 4, 0, sa, 0, sb);
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+            int row = (r & 3) + 8 * (r >> 2) + 4 * kgrp;
+            Dout[L * 1024 + row * 32 + col] = acc[r];
+        }
+    }
+}
+
 // Register-resident fp4 issue-rate ubench (4 independent f32x16 accumulators)
 extern "C" __global__ void __launch_bounds__(PROBE_BLOCK)
 mfma_fp4_loop_kernel(const int* __restrict__ seed, float* __restrict__ sink,

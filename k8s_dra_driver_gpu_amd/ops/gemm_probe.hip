@@ -2118,6 +2118,164 @@ gemm_fp4_3buf_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt, fl
     gemm_fp4_3buf_body<16, 1>(A, Bt, C, M, N, K);
 }
 
+// MX-scaled fp4 GEMM: real per-block E8M0 scales. The fp4 scale-lane
+// layout is the NAIVE one (probe profiles/r2/fp4_scale_probe.txt: scale
+// lane idx+32*g covers exactly the 32 k-elements that lane supplies), so
+// no chunk permutation is needed; scale panels are staged into LDS per
+// buffer exactly like the fp8 556 variant. blk = s*4 + d*2 + kgrp.
+template <int GROUP, int SWZ, int TBM = 256, int TBN = 256,
+          int TWR = 4, int TWC = 2>
+__device__ __forceinline__ void gemm_fp4_scaled_body(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
+    const unsigned char* __restrict__ SA, const unsigned char* __restrict__ SBt,
+    float* __restrict__ C, int M, int N, int K) {
+    constexpr int BK = 128, NT = TWR * TWC * WAVE;
+    constexpr int AI = (TBM / TWR) / 32, BJ = (TBN / TWC) / 32;
+    constexpr int RB = BK / 2;  // 64 B rows
+    constexpr int SCB = (TBM + TBN) * 4;  // per-buffer scale panel
+    __shared__ unsigned char lds[3 * ((TBM + TBN) * RB + SCB)];
+    const int HALF = (TBM + TBN) * RB + SCB;
+    auto ldsA = [&](int buf) -> unsigned char* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> unsigned char* { return lds + buf * HALF + TBM * RB; };
+    auto ldsS = [&](int buf) -> unsigned char* { return lds + buf * HALF + (TBM + TBN) * RB; };
+
+    const int tiles_n = (N + TBN - 1) / TBN;
+    int tile_m, tile_n;
+    if (GROUP > 1) {
+        const int tiles_m = (M + TBM - 1) / TBM;
+        const int per_group = GROUP * tiles_n;
+        const int gid = blockIdx.x / per_group;
+        const int first_m = gid * GROUP;
+        const int gsz = min(GROUP, tiles_m - first_m);
+        tile_m = first_m + (blockIdx.x % per_group) % gsz;
+        tile_n = (blockIdx.x % per_group) / gsz;
+    } else {
+        tile_m = blockIdx.x / tiles_n;
+        tile_n = blockIdx.x % tiles_n;
+    }
+    const int m0 = tile_m * TBM;
+    const int n0 = tile_n * TBN;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid / TWC) * (TBM / TWR);
+    const int wc = (wid % TWC) * (TBN / TWC);
+    const int kblocks_ = K / 32;
+
+    constexpr int ACH = TBM * (RB / 16);
+    constexpr int APH = ACH / NT;
+    constexpr int BCH = TBN * (RB / 16);
+    constexpr int BPH = BCH / NT;
+    static_assert(APH * NT == ACH && BPH * NT == BCH, "phase split");
+    constexpr int S = APH + BPH;
+    auto stage = [&](int buf, int k0) {
+        unsigned char* la = ldsA(buf);
+        unsigned char* lb = ldsB(buf);
+        const size_t kb0 = (size_t)k0 / 2;
+#pragma unroll
+        for (int phase = 0; phase < APH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (RB / 16);
+            int c = chunk % (RB / 16);
+            if (SWZ) c ^= r & (RB / 16 - 1);
+            const unsigned char* ga = &A[(size_t)(m0 + r) * (K / 2) + kb0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+#pragma unroll
+        for (int phase = 0; phase < BPH; ++phase) {
+            int chunk = phase * NT + tid;
+            int r = chunk / (RB / 16);
+            int c = chunk % (RB / 16);
+            if (SWZ) c ^= r & (RB / 16 - 1);
+            const unsigned char* gb = &Bt[(size_t)(n0 + r) * (K / 2) + kb0 + c * 16];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 16),
+                16, 0, 0);
+        }
+        if (tid < TBM + TBN) {
+            const unsigned char* gs = tid < TBM
+                ? &SA[(size_t)(m0 + tid) * kblocks_ + k0 / 32]
+                : &SBt[(size_t)(n0 + (tid - TBM)) * kblocks_ + k0 / 32];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gs,
+                (__attribute__((address_space(3))) void*)(ldsS(buf) + tid * 4),
+                4, 0, 0);
+        }
+    };
+
+    f32x16 acc[AI][BJ] = {};
+    const int kgrp = lane >> 5;
+    const int ln31 = lane & 31;
+    const int steps = K / BK;
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s % 3;
+        if (s + 1 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 2 < steps) stage((s + 2) % 3, (s + 2) * BK);
+
+        const unsigned char* la = ldsA(buf);
+        const unsigned char* lb = ldsB(buf);
+        const unsigned char* lscale = ldsS(buf);
+#pragma unroll
+        for (int d = 0; d < BK / 64; ++d) {
+            union { i32x8 v; unsigned char b[32]; } af[AI], bf[BJ];
+            int sa[AI], sb[BJ];
+#pragma unroll
+            for (int i = 0; i < AI; ++i) {
+                const int ar = wr + i * 32 + ln31;
+                int ch = d * 2 + kgrp;
+                if (SWZ) ch ^= ar & (RB / 16 - 1);
+                *(i32x4*)&af[i].b[0] = *(const i32x4*)&la[ar * RB + ch * 16];
+                sa[i] = lscale[ar * 4 + d * 2 + kgrp];
+            }
+#pragma unroll
+            for (int j = 0; j < BJ; ++j) {
+                const int bc = wc + j * 32 + ln31;
+                int ch = d * 2 + kgrp;
+                if (SWZ) ch ^= bc & (RB / 16 - 1);
+                *(i32x4*)&bf[j].b[0] = *(const i32x4*)&lb[bc * RB + ch * 16];
+                sb[j] = lscale[TBM * 4 + bc * 4 + d * 2 + kgrp];
+            }
+#pragma unroll
+            for (int i = 0; i < AI; ++i)
+#pragma unroll
+                for (int j = 0; j < BJ; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+                        af[i].v, bf[j].v, acc[i][j], 4, 4, 0, sa[i], 0, sb[j]);
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < AI; ++i)
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                int row = m0 + wr + i * 32 + (r & 3) + 8 * (r >> 2) + 4 * kgrp;
+                int col = n0 + wc + j * 32 + ln31;
+                if (row < M && col < N) C[(size_t)row * N + col] = acc[i][j][r];
+            }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_fp4_scaled_g16_swz_kernel(const unsigned char* A, const unsigned char* Bt,
+                               const unsigned char* SA, const unsigned char* SBt,
+                               float* C, int M, int N, int K) {
+    gemm_fp4_scaled_body<16, 1>(A, Bt, SA, SBt, C, M, N, K);
+}
+
 // 512x256 tile: 16 waves (1024 threads), 3 x 48 KiB LDS. MEASURED
 // CATASTROPHIC (273/169 TF vs 3524/2944 for the 256x256 3-buf champion,
 // numerics exact — gpurun_out/r2s29): the 1024-thread launch bounds force

@@ -45,6 +45,9 @@ extern "C" __global__ void mfma_fp4_loop_kernel(const int*, float*, int);
 extern "C" __global__ void mfma_fp4_tile_gemm_kernel(const unsigned char*, const unsigned char*, float*, int);
 extern "C" __global__ void mfma_fp8_scaled_tile_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, float*, int);
 extern "C" __global__ void mfma_scale_probe_kernel(float*);
+extern "C" __global__ void mfma_fp4_scale_probe_kernel(float*);
+extern "C" __global__ void mfma_fp4_scaled_tile_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, float*, int);
+extern "C" __global__ void gemm_fp4_scaled_g16_swz_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, float*, int, int, int);
 extern "C" __global__ void mfma_bf16_tile_gemm_kernel(const short*, const short*, float*, int);
 extern "C" __global__ void mfma_fp8_tile_gemm_kernel(const unsigned char*, const unsigned char*, float*, int, int);
 extern "C" __global__ void p2p_read_kernel(float4v*, const float4v*, long);
@@ -698,6 +701,18 @@ int fp_mfma_scale_probe_host(int dev, float* D) {
     return 0;
 }
 
+int fp_mfma_fp4_scale_probe_host(int dev, float* D) {
+    CHKI(hipSetDevice(dev));
+    float* dD;
+    CHKI(hipMalloc(&dD, 129 * 1024 * sizeof(float)));
+    hipLaunchKernelGGL(mfma_fp4_scale_probe_kernel, dim3(1), dim3(64), 0, 0, dD);
+    CHKI(hipGetLastError());
+    CHKI(hipDeviceSynchronize());
+    CHKI(hipMemcpy(D, dD, 129 * 1024 * sizeof(float), hipMemcpyDeviceToHost));
+    hipFree(dD);
+    return 0;
+}
+
 int fp_mfma_fp8_scaled_tile_host(int dev, const unsigned char* A, const unsigned char* B,
                                  const unsigned char* SA, const unsigned char* SB,
                                  float* D, int K) {
@@ -790,6 +805,90 @@ double fp_gemm_fp8_scaled_tflops(int dev, int size, int iters, int variant) {
     CHK(hipEventRecord(t0));
     for (int i = 0; i < iters; ++i)
         hipLaunchKernelGGL(kern, grid, dim3(threads), 0, 0, A, Bt, SA, SBt, C, M, N, K);
+    CHK(hipEventRecord(t1));
+    CHK(hipEventSynchronize(t1));
+    double ms = time_kernel_ms(t0, t1);
+    hipFree(A); hipFree(Bt); hipFree(SA); hipFree(SBt); hipFree(C);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    return 2.0 * M * (double)N * K * iters / (ms * 1e9);
+}
+
+int fp_mfma_fp4_scaled_tile_host(int dev, const unsigned char* A, const unsigned char* B,
+                                 const unsigned char* SA, const unsigned char* SB,
+                                 float* D, int K) {
+    CHKI(hipSetDevice(dev));
+    unsigned char *dA, *dB, *dSA, *dSB;
+    float* dD;
+    CHKI(hipMalloc(&dA, (size_t)32 * K / 2));
+    CHKI(hipMalloc(&dB, (size_t)K / 2 * 32));
+    CHKI(hipMalloc(&dSA, (size_t)32 * K / 32));
+    CHKI(hipMalloc(&dSB, (size_t)32 * K / 32));
+    CHKI(hipMalloc(&dD, 32 * 32 * sizeof(float)));
+    CHKI(hipMemcpy(dA, A, (size_t)32 * K / 2, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dB, B, (size_t)K / 2 * 32, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dSA, SA, (size_t)32 * K / 32, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dSB, SB, (size_t)32 * K / 32, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(mfma_fp4_scaled_tile_kernel, dim3(1), dim3(64), 0, 0,
+                       dA, dB, dSA, dSB, dD, K);
+    CHKI(hipGetLastError());
+    CHKI(hipDeviceSynchronize());
+    CHKI(hipMemcpy(D, dD, 32 * 32 * sizeof(float), hipMemcpyDeviceToHost));
+    hipFree(dA); hipFree(dB); hipFree(dSA); hipFree(dSB); hipFree(dD);
+    return 0;
+}
+
+int fp_gemm_fp4_scaled_host(int dev, const unsigned char* A, const unsigned char* Bt,
+                            const unsigned char* SA, const unsigned char* SBt,
+                            float* C, int M, int N, int K) {
+    CHKI(hipSetDevice(dev));
+    unsigned char *dA, *dB, *dSA, *dSB;
+    float* dC;
+    CHKI(hipMalloc(&dA, (size_t)M * K / 2));
+    CHKI(hipMalloc(&dB, (size_t)N * K / 2));
+    CHKI(hipMalloc(&dSA, (size_t)M * K / 32));
+    CHKI(hipMalloc(&dSB, (size_t)N * K / 32));
+    CHKI(hipMalloc(&dC, (size_t)M * N * sizeof(float)));
+    CHKI(hipMemcpy(dA, A, (size_t)M * K / 2, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dB, Bt, (size_t)N * K / 2, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dSA, SA, (size_t)M * K / 32, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dSB, SBt, (size_t)N * K / 32, hipMemcpyHostToDevice));
+    dim3 grid(((M + 255) / 256) * ((N + 255) / 256));
+    hipLaunchKernelGGL(gemm_fp4_scaled_g16_swz_kernel, grid, dim3(512), 0, 0,
+                       dA, dB, dSA, dSB, dC, M, N, K);
+    CHKI(hipGetLastError());
+    CHKI(hipDeviceSynchronize());
+    CHKI(hipMemcpy(C, dC, (size_t)M * N * sizeof(float), hipMemcpyDeviceToHost));
+    hipFree(dA); hipFree(dB); hipFree(dSA); hipFree(dSB); hipFree(dC);
+    return 0;
+}
+
+double fp_gemm_fp4_scaled_tflops(int dev, int size, int iters) {
+    CHK(hipSetDevice(dev));
+    int M = size, N = size, K = size;
+    unsigned char *A, *Bt, *SA, *SBt;
+    float* C;
+    CHK(hipMalloc(&A, (size_t)M * K / 2));
+    CHK(hipMalloc(&Bt, (size_t)N * K / 2));
+    CHK(hipMalloc(&SA, (size_t)M * K / 32));
+    CHK(hipMalloc(&SBt, (size_t)N * K / 32));
+    CHK(hipMalloc(&C, (size_t)M * N * sizeof(float)));
+    CHK(hipMemset(A, 0x22, (size_t)M * K / 2));
+    CHK(hipMemset(Bt, 0x11, (size_t)N * K / 2));
+    CHK(hipMemset(SA, 0x7F, (size_t)M * K / 32));
+    CHK(hipMemset(SBt, 0x7F, (size_t)N * K / 32));
+    dim3 grid((M / 256) * (N / 256));
+    hipEvent_t t0, t1;
+    CHK(hipEventCreate(&t0));
+    CHK(hipEventCreate(&t1));
+    hipLaunchKernelGGL(gemm_fp4_scaled_g16_swz_kernel, grid, dim3(512), 0, 0,
+                       A, Bt, SA, SBt, C, M, N, K);
+    CHK(hipGetLastError());
+    CHK(hipDeviceSynchronize());
+    CHK(hipEventRecord(t0));
+    for (int i = 0; i < iters; ++i)
+        hipLaunchKernelGGL(gemm_fp4_scaled_g16_swz_kernel, grid, dim3(512), 0, 0,
+                           A, Bt, SA, SBt, C, M, N, K);
     CHK(hipEventRecord(t1));
     CHK(hipEventSynchronize(t1));
     double ms = time_kernel_ms(t0, t1);

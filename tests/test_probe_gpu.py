@@ -386,6 +386,32 @@ class TestFp8Probe:
             err = np.abs(c - ref).max() / np.abs(ref).max()
             assert err < 1e-4, f"variant {v}: {err}"
 
+    def test_mx_scaled_fp4_numerics(self, probe):
+        # fp4 (e2m1) with real per-block E8M0 scales: tile + full GEMM vs
+        # the dequantized MX reference. The fp4 scale layout is the naive
+        # one (each lane scales its own 32-k block) — pinned by
+        # mfma_fp4_scale_probe_kernel on hardware.
+        rng = np.random.default_rng(44)
+        a = (rng.standard_normal((32, 256))
+             * np.exp2(rng.integers(-6, 7, (32, 256)))).astype(np.float32)
+        b = (rng.standard_normal((256, 32))
+             * np.exp2(rng.integers(-6, 7, (256, 32)))).astype(np.float32)
+        d = probe.mfma_fp4_scaled_tile(a, b)
+        a4, sa = probe.mx_quantize_fp4(a)
+        b4t, sb = probe.mx_quantize_fp4(np.ascontiguousarray(b.T))
+        ref = (probe.mx_dequantize_fp4(a4, sa).astype(np.float64)
+               @ probe.mx_dequantize_fp4(b4t, sb).astype(np.float64).T)
+        assert np.abs(d - ref).max() / np.abs(ref).max() < 1e-4
+        M = N = K = 512
+        ag = (rng.standard_normal((M, K))
+              * np.exp2(rng.integers(-6, 7, (M, K)))).astype(np.float32)
+        btg = (rng.standard_normal((N, K))
+               * np.exp2(rng.integers(-6, 7, (N, K)))).astype(np.float32)
+        c, a4, sa, b4t, sbt = probe.gemm_fp4_scaled(ag, btg)
+        ref = (probe.mx_dequantize_fp4(a4, sa).astype(np.float64)
+               @ probe.mx_dequantize_fp4(b4t, sbt).astype(np.float64).T)
+        assert np.abs(c - ref).max() / np.abs(ref).max() < 1e-4
+
     def test_fp8_codec_round_trip(self, probe):
         # CPU-only property of the host codec, kept here with the fp8 suite
         rng = np.random.default_rng(0)
