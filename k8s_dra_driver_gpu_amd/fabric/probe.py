@@ -692,6 +692,10 @@ class FabricProbeReport:
     # low-precision matrix-core floors (MX-scaled fp8/fp4; 0.0 = not run)
     mfma_fp8_tflops: float = 0.0
     mfma_fp4_tflops: float = 0.0
+    # quantized-GEMM floors with REAL per-block E8M0 scales flowing
+    # through the mfma scale operands (the production MX inference path)
+    gemm_fp8_mx_tflops: float = 0.0
+    gemm_fp4_mx_tflops: float = 0.0
 
 
 def run_fabric_report(quick: bool = True) -> FabricProbeReport:
@@ -729,4 +733,6 @@ def run_fabric_report(quick: bool = True) -> FabricProbeReport:
         allreduce_gbps=ar,
         mfma_fp8_tflops=mfma_fp8_tflops(0, 1024, 5),
         mfma_fp4_tflops=mfma_fp4_tflops(0, 1024, 5),
+        gemm_fp8_mx_tflops=gemm_fp8_scaled_tflops(0, 2048 if quick else 8192, 5),
+        gemm_fp4_mx_tflops=gemm_fp4_scaled_tflops(0, 2048 if quick else 8192, 5),
     )
