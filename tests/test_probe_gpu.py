@@ -384,7 +384,11 @@ class TestFp8Probe:
             ref = (probe.mx_dequantize_fp8(a8, sa).astype(np.float64)
                    @ probe.mx_dequantize_fp8(b8t, sbt).astype(np.float64).T)
             err = np.abs(c - ref).max() / np.abs(ref).max()
-            assert err < 1e-4, f"variant {v}: {err}"
+            # fp32 mfma accumulation vs the f64 reference at +/-2^6
+            # per-element dynamic range sits near 1e-4; a wrong scale
+            # mapping fails at ~1 (measured 0.98), so 5e-4 separates
+            # cleanly without seed sensitivity
+            assert err < 5e-4, f"variant {v}: {err}"
 
     def test_mx_scaled_fp4_numerics(self, probe):
         # fp4 (e2m1) with real per-block E8M0 scales: tile + full GEMM vs
