@@ -2228,6 +2228,16 @@ __device__ __forceinline__ void gemm_fp4_scaled_body(
         const unsigned char* la = ldsA(buf);
         const unsigned char* lb = ldsB(buf);
         const unsigned char* lscale = ldsS(buf);
+        // one b32 read per row fetches all 4 of this step's scale bytes;
+        // both d-iterations extract theirs with a shift (halves the LDS
+        // scale traffic vs per-(d,row) byte reads)
+        unsigned sa32[AI], sb32[BJ];
+#pragma unroll
+        for (int i = 0; i < AI; ++i)
+            sa32[i] = *(const unsigned*)&lscale[(wr + i * 32 + ln31) * 4];
+#pragma unroll
+        for (int j = 0; j < BJ; ++j)
+            sb32[j] = *(const unsigned*)&lscale[TBM * 4 + (wc + j * 32 + ln31) * 4];
 #pragma unroll
         for (int d = 0; d < BK / 64; ++d) {
             union { i32x8 v; unsigned char b[32]; } af[AI], bf[BJ];
@@ -2238,7 +2248,7 @@ __device__ __forceinline__ void gemm_fp4_scaled_body(
                 int ch = d * 2 + kgrp;
                 if (SWZ) ch ^= ar & (RB / 16 - 1);
                 *(i32x4*)&af[i].b[0] = *(const i32x4*)&la[ar * RB + ch * 16];
-                sa[i] = lscale[ar * 4 + d * 2 + kgrp];
+                sa[i] = (sa32[i] >> (8 * (d * 2 + kgrp))) & 0xFF;
             }
 #pragma unroll
             for (int j = 0; j < BJ; ++j) {
@@ -2246,7 +2256,7 @@ __device__ __forceinline__ void gemm_fp4_scaled_body(
                 int ch = d * 2 + kgrp;
                 if (SWZ) ch ^= bc & (RB / 16 - 1);
                 *(i32x4*)&bf[j].b[0] = *(const i32x4*)&lb[bc * RB + ch * 16];
-                sb[j] = lscale[TBM * 4 + bc * 4 + d * 2 + kgrp];
+                sb[j] = (sb32[j] >> (8 * (d * 2 + kgrp))) & 0xFF;
             }
 #pragma unroll
             for (int i = 0; i < AI; ++i)
