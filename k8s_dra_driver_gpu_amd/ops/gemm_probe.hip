@@ -1375,7 +1375,10 @@ __device__ __forceinline__ void gemm_fp8_scaled_body(
         if (LOADS == 2 && tid < TBM + TBN) {
             // this k-step's 4 scale bytes for one A row / B column. A
             // masked instruction still bumps vmcnt for waves with any
-            // active lane, which only makes the vmcnt(S) waits stricter.
+            // active lane, which only makes the vmcnt(S) waits stricter
+            // (measured FASTER than the unmasked/padded alternative for
+            // this shape: 1837 vs 1736 TF — the fp4 body keeps the
+            // unmasked form, where it won +5%).
             const unsigned char* gs = tid < TBM
                 ? &SA[(size_t)(m0 + tid) * kblocks_ + k0 / 32]
                 : &SBt[(size_t)(n0 + (tid - TBM)) * kblocks_ + k0 / 32];
@@ -1583,7 +1586,7 @@ __device__ __forceinline__ void gemm_fp8_scaled4_body(
         for (int ss = 0; ss < 4; ++ss) {
             const int s = s4 + ss;
             const int buf = s % 3;
-            // strict count: the 8 group-start scale loads sit between
+            // (4-step variant unchanged) strict count: scale loads sit between
             // stage(s4+1) and stage(s4+2) in issue order, so vmcnt(S)
             // remains sufficient at every inner step (at ss=0/1 it also
             // drains the scale loads — one L2 latency per 4 steps)
@@ -2132,7 +2135,7 @@ __device__ __forceinline__ void gemm_fp4_scaled_body(
     constexpr int BK = 128, NT = TWR * TWC * WAVE;
     constexpr int AI = (TBM / TWR) / 32, BJ = (TBN / TWC) / 32;
     constexpr int RB = BK / 2;  // 64 B rows
-    constexpr int SCB = (TBM + TBN) * 4;  // per-buffer scale panel
+    constexpr int SCB = NT * 4;  // per-buffer scale panel (thread-padded)
     __shared__ unsigned char lds[3 * ((TBM + TBN) * RB + SCB)];
     const int HALF = (TBM + TBN) * RB + SCB;
     auto ldsA = [&](int buf) -> unsigned char* { return lds + buf * HALF; };
@@ -2197,10 +2200,12 @@ __device__ __forceinline__ void gemm_fp4_scaled_body(
                 (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * NT + wid * WAVE) * 16),
                 16, 0, 0);
         }
-        if (tid < TBM + TBN) {
-            const unsigned char* gs = tid < TBM
-                ? &SA[(size_t)(m0 + tid) * kblocks_ + k0 / 32]
-                : &SBt[(size_t)(n0 + (tid - TBM)) * kblocks_ + k0 / 32];
+        {
+            // unmasked (clamped) so every wave issues exactly S+1 loads
+            const int t = tid < TBM + TBN - 1 ? tid : TBM + TBN - 1;
+            const unsigned char* gs = t < TBM
+                ? &SA[(size_t)(m0 + t) * kblocks_ + k0 / 32]
+                : &SBt[(size_t)(n0 + (t - TBM)) * kblocks_ + k0 / 32];
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) void*)gs,
                 (__attribute__((address_space(3))) void*)(ldsS(buf) + tid * 4),
@@ -2218,7 +2223,7 @@ __device__ __forceinline__ void gemm_fp4_scaled_body(
     for (int s = 0; s < steps; ++s) {
         const int buf = s % 3;
         if (s + 1 < steps)
-            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S + 1) : "memory");
         else
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");

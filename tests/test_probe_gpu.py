@@ -416,6 +416,15 @@ class TestFp8Probe:
                @ probe.mx_dequantize_fp4(b4t, sbt).astype(np.float64).T)
         assert np.abs(c - ref).max() / np.abs(ref).max() < 1e-4
 
+    def test_mx_scaled_throughput_floor(self, probe):
+        # regression floor for the production MX path (measured 1678/2529
+        # TF at 4096^3 on the r2 boxes; floor at ~60% of that)
+        tf8 = probe.gemm_fp8_scaled_tflops(size=4096, iters=5)
+        tf4 = probe.gemm_fp4_scaled_tflops(size=4096, iters=5)
+        print(f"\nMX-scaled GEMM: fp8 {tf8:.0f} TF, fp4 {tf4:.0f} TF @4096^3")
+        assert tf8 > 1000, f"fp8 MX GEMM too slow: {tf8:.0f} TF"
+        assert tf4 > 1500, f"fp4 MX GEMM too slow: {tf4:.0f} TF"
+
     def test_fp8_codec_round_trip(self, probe):
         # CPU-only property of the host codec, kept here with the fp8 suite
         rng = np.random.default_rng(0)
