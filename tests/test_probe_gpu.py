@@ -199,6 +199,39 @@ class TestFullStackRealGpu:
         finally:
             cluster.stop()
 
+    def test_shared_gpu_claim_on_real_gpu(self, tmp_path):
+        # BASELINE config 3 on hardware: one claim shared by two containers
+        # of a pod (gpu-test2) — TimeSlicing config flows into the CDI
+        # spec's env edits and the real device nodes are injected once.
+        import glob as _glob
+        import json as _json
+
+        from k8s_dra_driver_gpu_amd.bench.localcluster import LocalCluster
+
+        cluster = LocalCluster(
+            real_devices=True, work_dir=str(tmp_path), partitionable=False
+        ).start()
+        try:
+            specs = os.path.join(
+                os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                "demo", "specs", "quickstart",
+            )
+            ev = cluster.apply_yaml(os.path.join(specs, "gpu-test2.yaml"))
+            assert any("prepared gpu-" in e for e in ev), ev
+            spec_files = _glob.glob(os.path.join(str(tmp_path), "cdi", "*claim*.json"))
+            assert spec_files, "no per-claim CDI spec written"
+            envs, paths = [], []
+            for f in spec_files:
+                spec = _json.load(open(f))
+                for d in spec["devices"]:
+                    ce = d.get("containerEdits", {})
+                    envs += ce.get("env", []) or []
+                    paths += [n["path"] for n in ce.get("deviceNodes", []) or []]
+            assert "/dev/kfd" in paths
+            assert any(e.startswith("AMDDRA_SHARING=") for e in envs), envs
+        finally:
+            cluster.stop()
+
 
 class TestBurnDiagnostic:
     def test_concurrent_mfma_hbm_burn(self, probe):
