@@ -231,6 +231,16 @@ def main() -> None:
 
         gbps = probe.hbm_read_gbps(local_rank % max(1, probe.device_count()), 1 << 30, 3)
         fabric["hbm_read_gbps"] = round(gbps, 1)
+        if rank == 0:
+            # MX quantized-GEMM floors (real per-block E8M0 scales through
+            # the mfma scale operands) — the production low-precision path
+            try:
+                fabric["gemm_fp8_mx_tflops"] = round(
+                    probe.gemm_fp8_scaled_tflops(0, 4096, 3), 1)
+                fabric["gemm_fp4_mx_tflops"] = round(
+                    probe.gemm_fp4_scaled_tflops(0, 4096, 3), 1)
+            except Exception:
+                pass
         if rank == 0 and probe.device_count() > 1:
             # one xGMI link-pair measurement for the record (per-link ~153 GB/s
             # x links between the pair; full matrix via fabric.probe)

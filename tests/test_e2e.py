@@ -390,11 +390,41 @@ class TestLocalClusterDemo:
             assert len([e for e in ev6 if "prepared gpu-" in e]) == 2, ev6
             ev5 = cluster.apply_yaml(os.path.join(specs, "gpu-test-extres.yaml"))
             assert any("(extended-resource)" in e for e in ev5), ev5
+            ev2 = cluster.apply_yaml(os.path.join(specs, "gpu-test2.yaml"))
+            assert any("prepared gpu-" in e for e in ev2), ev2
+            ev7 = cluster.apply_yaml(os.path.join(specs, "gpu-test5.yaml"))
+            assert any("prepared gpu-" in e for e in ev7), ev7
             ev4 = cluster.apply_yaml(os.path.join(specs, "cd-test1.yaml"))
             assert cluster.wait_cd_ready("cd1", "cd-test1"), ev4
             assert any("prepared channel-0" in e for e in ev4), ev4
             # pod deletion unprepares and frees the devices
             cluster.delete_pod("gpu-test1", "pod1")
+        finally:
+            cluster.stop()
+
+    def test_quickstart_vfio_spec(self, tmp_path):
+        # gpu-test-vfio end-to-end on the mock vfio manager: the claim
+        # rebinds the device and the CDI spec injects the vfio group node
+        from k8s_dra_driver_gpu_amd.bench.localcluster import LocalCluster
+
+        specs = os.path.join(
+            os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+            "demo", "specs", "quickstart",
+        )
+        cluster = LocalCluster(num_gpus=2, work_dir=str(tmp_path),
+                               vfio=True).start()
+        try:
+            ev = cluster.apply_yaml(os.path.join(specs, "gpu-test-vfio.yaml"))
+            assert any("prepared" in e for e in ev), ev
+            import glob as _glob
+            import json as _json
+            found = []
+            for f in _glob.glob(os.path.join(str(tmp_path), "cdi", "*claim*.json")):
+                spec = _json.load(open(f))
+                for d in spec["devices"]:
+                    found += [n["path"] for n in
+                              d.get("containerEdits", {}).get("deviceNodes", []) or []]
+            assert any("/dev/vfio/" in p for p in found), found
         finally:
             cluster.stop()
 
