@@ -297,3 +297,41 @@ class TestPartitionNameProperties:
             parse_partition_name(s)
         except ValueError:
             pass
+
+
+class TestCheckpointFragmentCache:
+    """The checkpoint's canonical payload is composed from per-claim cached
+    JSON fragments (plugin/checkpoint.py canonical_payload). The invariant:
+    after ANY sequence of set/remove operations, the composed payload is
+    byte-identical to serializing the whole object from scratch — a stale
+    or mis-formatted fragment would silently corrupt checksums across
+    driver restarts."""
+
+    @given(ops=st.lists(
+        st.tuples(st.sampled_from(["set", "remove", "reset"]),
+                  st.integers(0, 7)),
+        min_size=1, max_size=24))
+    @settings(max_examples=60, deadline=None)
+    def test_fragment_composition_matches_full_serialization(self, ops):
+        from k8s_dra_driver_gpu_amd.api.serde import to_dict
+        from k8s_dra_driver_gpu_amd.plugin.checkpoint import (
+            CheckpointData, ClaimRef, PreparedClaim, PreparedDevice,
+            _canonical,
+        )
+
+        data = CheckpointData(node_boot_id="boot-1")
+        for i, (op, n) in enumerate(ops):
+            uid = f"uid-{n}"
+            if op == "set":
+                data.set_claim(uid, PreparedClaim(
+                    state="PrepareCompleted" if i % 2 else "PrepareStarted",
+                    claim=ClaimRef("ns", f"c{i}", uid),
+                    devices=[PreparedDevice(type="gpu", name=f"gpu-{i}")]))
+            elif op == "remove":
+                data.remove_claim(uid)
+            else:
+                # simulate a reload: fragments rebuilt lazily from raw dicts
+                data = CheckpointData(
+                    node_boot_id=data.node_boot_id,
+                    prepared_claims=dict(data.prepared_claims))
+            assert data.canonical_payload() == _canonical(to_dict(data))
