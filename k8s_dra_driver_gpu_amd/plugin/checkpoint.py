@@ -256,7 +256,10 @@ class CheckpointManager:
             f.write(raw_s)
             if durable:
                 f.flush()
-                os.fsync(f.fileno())
+                # fdatasync: the claim payload must be durable before we
+                # touch devices; the inode metadata (mtime) need not be.
+                # os.replace below gives atomic visibility either way.
+                os.fdatasync(f.fileno())
         os.replace(tmp, self.path)
         self._cache_payload = payload_s
         self._cache_stat = self._stat_key()
