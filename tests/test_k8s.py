@@ -615,3 +615,75 @@ class TestWatchCacheSemantics:
             time.sleep(0.05)
         assert int(inf.last_resource_version) > rv0
         inf.stop()
+
+
+class TestInformerRandomizedConvergence:
+    """Randomized model test: drive the fake API server through hundreds of
+    create/update/delete ops with an aggressively small watch-cache history
+    (so the informer repeatedly hits 410 Gone and must relist), verifying
+    at interleaved checkpoints that the informer cache converges exactly to
+    the server's state. Seeded for reproducibility."""
+
+    def _converged(self, inf, server, timeout=5.0):
+        import time as _t
+
+        deadline = _t.monotonic() + timeout
+        want = None
+        while _t.monotonic() < deadline:
+            objs = server.list("computedomains")
+            want = {
+                f"{o['metadata'].get('namespace','')}/{o['metadata']['name']}":
+                    o["metadata"]["resourceVersion"]
+                for o in objs
+            }
+            have = {
+                f"{o['metadata'].get('namespace','')}/{o['metadata']['name']}":
+                    o["metadata"]["resourceVersion"]
+                for o in inf.items()
+            }
+            if have == want:
+                return True
+            _t.sleep(0.02)
+        return False
+
+    def test_converges_under_eviction_pressure(self):
+        import random
+
+        from k8s_dra_driver_gpu_amd.k8s.fakeserver import FakeApiServer
+        from k8s_dra_driver_gpu_amd.k8s.informer import Informer
+
+        rng = random.Random(1234)
+        server = FakeApiServer(history_limit=4)
+        inf = Informer(server, "computedomains").start()
+        try:
+            assert inf.wait_for_sync(5.0)
+            live = {}
+            for step in range(300):
+                op = rng.random()
+                if op < 0.45 or not live:
+                    name = f"cd-{rng.randrange(40)}"
+                    if name not in live:
+                        obj = {
+                            "apiVersion": "resource.amd.com/v1beta1",
+                            "kind": "ComputeDomain",
+                            "metadata": {"name": name, "namespace": "default"},
+                            "spec": {"numNodes": 1},
+                        }
+                        live[name] = server.create("computedomains", obj)
+                elif op < 0.75:
+                    name = rng.choice(list(live))
+                    cur = server.get("computedomains", name, "default")
+                    cur["spec"]["numNodes"] = rng.randrange(1, 9)
+                    live[name] = server.update("computedomains", cur)
+                else:
+                    name = rng.choice(list(live))
+                    server.delete("computedomains", name, "default")
+                    del live[name]
+                if step in (75, 150, 225):
+                    assert self._converged(inf, server), (
+                        f"informer diverged at step {step}"
+                    )
+            assert self._converged(inf, server), "informer diverged at end"
+            assert len(inf.items()) == len(live)
+        finally:
+            inf.stop()
