@@ -540,3 +540,57 @@ class TestHttpBookmarks:
         assert ev is not None and ev.type == "BOOKMARK", ev
         assert ev.object["metadata"]["resourceVersion"] == rv
         w.stop()
+
+
+class TestHttpInformerRandomizedConvergence:
+    """The randomized watch-cache convergence test over the REAL HTTP path
+    (MiniApiServer + HttpClient + Informer): mutations through the HTTP
+    client, tiny watch-cache history (eviction -> 410 Gone -> relist), and
+    a 2 s forced stream recycle via AMDDRA_WATCH_TIMEOUT."""
+
+    def test_converges_under_eviction_and_recycling(self, monkeypatch):
+        import random
+
+        from k8s_dra_driver_gpu_amd.k8s.fakeserver import FakeApiServer
+
+        monkeypatch.setenv("AMDDRA_WATCH_TIMEOUT", "1")
+        srv = MiniApiServer(api=FakeApiServer(history_limit=4))
+        srv.start()
+        c = HttpClient(base_url=f"http://127.0.0.1:{srv.port}")
+        inf = Informer(c, "computedomains").start()
+        try:
+            assert inf.wait_for_sync(5.0)
+            rng = random.Random(99)
+            live = set()
+            for step in range(60):
+                op = rng.random()
+                if op < 0.5 or not live:
+                    name = f"cd-{rng.randrange(25)}"
+                    if name not in live:
+                        c.create("computedomains", cd(name))
+                        live.add(name)
+                elif op < 0.75:
+                    name = rng.choice(sorted(live))
+                    cur = c.get("computedomains", name, "default")
+                    cur["spec"]["numNodes"] = rng.randrange(1, 9)
+                    c.update("computedomains", cur)
+                else:
+                    name = rng.choice(sorted(live))
+                    c.delete("computedomains", name, "default")
+                    live.discard(name)
+                if step == 30:
+                    time.sleep(1.2)  # span at least one forced recycle
+            deadline = time.monotonic() + 8.0
+            while time.monotonic() < deadline:
+                want = {o["metadata"]["name"]: o["metadata"]["resourceVersion"]
+                        for o in c.list("computedomains")}
+                have = {o["metadata"]["name"]: o["metadata"]["resourceVersion"]
+                        for o in inf.items()}
+                if have == want:
+                    break
+                time.sleep(0.05)
+            assert have == want, f"diverged: {len(have)} vs {len(want)}"
+            assert sorted(have) == sorted(live)
+        finally:
+            inf.stop()
+            srv.stop()
