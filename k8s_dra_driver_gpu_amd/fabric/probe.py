@@ -642,6 +642,8 @@ def gemm_fp8_scaled(a: np.ndarray, bt: np.ndarray, dev: int = 0,
     N, K2 = bt.shape
     tile_m = 128 if variant == 5 else 256
     assert K == K2 and M % tile_m == 0 and N % 128 == 0 and K % 128 == 0
+    # the 4-step-grouped variant iterates K in 512-element groups
+    assert variant != 546 or K % 512 == 0
     a8, sa = mx_quantize_fp8(np.ascontiguousarray(a, dtype=np.float32))
     b8t, sbt = mx_quantize_fp8(np.ascontiguousarray(bt, dtype=np.float32))
     out = np.zeros((M, N), dtype=np.float32)
