@@ -97,6 +97,23 @@ def _load() -> ctypes.CDLL:
         ctypes.POINTER(ctypes.c_float),
         ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
     ]
+    lib.fp_gemm_bf16_splitk_tflops.restype = ctypes.c_double
+    lib.fp_gemm_bf16_splitk_tflops.argtypes = [
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+    ]
+    lib.fp_gemm_bf16_splitk_tflops_mnk.restype = ctypes.c_double
+    lib.fp_gemm_bf16_splitk_tflops_mnk.argtypes = [
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+        ctypes.c_int, ctypes.c_int,
+    ]
+    lib.fp_gemm_bf16_splitk_host.restype = ctypes.c_int
+    lib.fp_gemm_bf16_splitk_host.argtypes = [
+        ctypes.c_int,
+        ctypes.POINTER(ctypes.c_int16),
+        ctypes.POINTER(ctypes.c_int16),
+        ctypes.POINTER(ctypes.c_float),
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+    ]
     lib.fp_gemm_fp8_scaled_host.restype = ctypes.c_int
     lib.fp_gemm_fp8_scaled_host.argtypes = [
         ctypes.c_int,
@@ -666,6 +683,44 @@ def gemm_fp8_scaled_tflops(dev: int = 0, size: int = 4096, iters: int = 10,
     """MX-scaled fp8 GEMM throughput (real per-block scales in the loop)."""
     return _check(_load().fp_gemm_fp8_scaled_tflops(dev, size, iters, variant),
                   "gemm_fp8_scaled_tflops")
+
+
+def gemm_bf16_splitk(a: np.ndarray, bt: np.ndarray, ksplit: int = 4,
+                     dev: int = 0) -> np.ndarray:
+    """C[M,N] = a @ bt^T via the split-K 128x128 kernel: `ksplit` partial
+    products per tile accumulated with f32 hardware atomics — fills the
+    chip on shapes where plain tiling launches < 256 workgroups."""
+    M, K = a.shape
+    N, K2 = bt.shape
+    assert K == K2 and K % (32 * ksplit) == 0 and (K // ksplit) % 32 == 0
+    a16 = _to_bf16_bits(np.ascontiguousarray(a, dtype=np.float32))
+    b16 = _to_bf16_bits(np.ascontiguousarray(bt, dtype=np.float32))
+    out = np.zeros((M, N), dtype=np.float32)
+    rc = _load().fp_gemm_bf16_splitk_host(
+        dev,
+        a16.ctypes.data_as(ctypes.POINTER(ctypes.c_int16)),
+        b16.ctypes.data_as(ctypes.POINTER(ctypes.c_int16)),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+        M, N, K, ksplit,
+    )
+    if rc < 0:
+        raise ProbeError(f"gemm_bf16_splitk failed with hip error {-rc}")
+    return out
+
+
+def gemm_bf16_splitk_tflops(dev: int = 0, size: int = 2048, iters: int = 10,
+                            ksplit: int = 4) -> float:
+    """Split-K bf16 GEMM throughput (C zeroing included in each iteration)."""
+    return _check(_load().fp_gemm_bf16_splitk_tflops(dev, size, iters, ksplit),
+                  "gemm_bf16_splitk_tflops")
+
+
+def gemm_bf16_splitk_tflops_mnk(M: int, N: int, K: int, dev: int = 0,
+                                iters: int = 10, ksplit: int = 8) -> float:
+    """Split-K throughput on an explicit (M, N, K) — the kernel's target is
+    tall-skinny K (few output tiles, huge reduction dim)."""
+    return _check(_load().fp_gemm_bf16_splitk_tflops_mnk(
+        dev, M, N, K, iters, ksplit), "gemm_bf16_splitk_tflops_mnk")
 
 
 def _to_bf16_bits(x: np.ndarray) -> np.ndarray:

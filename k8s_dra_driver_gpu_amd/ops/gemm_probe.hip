@@ -497,6 +497,120 @@ __device__ __forceinline__ void gemm_bf16_128_pipe2_body(
     }
 }
 
+// Split-K variant of the depth-2 128x128 kernel for SMALL shapes: at
+// 2048^3 the plain kernel launches only 256 WGs (chip barely covered) and
+// 1024^3 only 64. Each workgroup computes a partial product over one of
+// `ksplit` K-slices and accumulates into C with hardware f32 atomic adds
+// (unsafeAtomicAdd -> global_atomic_add_f32); the host zeroes C first.
+// Requires (K/ksplit) % BK == 0.
+template <int BK>
+__device__ __forceinline__ void gemm_bf16_splitk_body(
+    const short* __restrict__ A, const short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int ksplit) {
+    __shared__ short lds[4 * (BM * BK + BN * BK)];
+    const int HALF = BM * BK + BN * BK;
+    auto ldsA = [&](int buf) -> short* { return lds + buf * HALF; };
+    auto ldsB = [&](int buf) -> short* { return lds + buf * HALF + BM * BK; };
+
+    const int tiles_n = (N + BN - 1) / BN;
+    const int tiles_mn = ((M + BM - 1) / BM) * tiles_n;
+    const int tile = blockIdx.x % tiles_mn;   // tile-major: slices of one
+    const int slice = blockIdx.x / tiles_mn;  // tile land on distinct CUs
+    const int tile_m = tile / tiles_n;
+    const int tile_n = tile % tiles_n;
+    const int m0 = tile_m * BM;
+    const int n0 = tile_n * BN;
+    const int kper = K / ksplit;
+    const int kbase = slice * kper;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int wr = (wid >> 1) * 64;
+    const int wc = (wid & 1) * 64;
+
+    constexpr int CHUNKS = BM * (BK / 8);
+    constexpr int PHASES = CHUNKS / THREADS;
+    constexpr int S = PHASES * 2;
+    auto stage = [&](int buf, int k0) {
+        short* la = ldsA(buf);
+        short* lb = ldsB(buf);
+#pragma unroll
+        for (int phase = 0; phase < PHASES; ++phase) {
+            int chunk = phase * THREADS + tid;
+            int r = chunk / (BK / 8);
+            int c = chunk % (BK / 8);
+            const short* ga = &A[(size_t)(m0 + r) * K + kbase + k0 + c * 8];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)ga,
+                (__attribute__((address_space(3))) void*)(la + (size_t)(phase * THREADS + wid * WAVE) * 8),
+                16, 0, 0);
+            const short* gb = &Bt[(size_t)(n0 + r) * K + kbase + k0 + c * 8];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gb,
+                (__attribute__((address_space(3))) void*)(lb + (size_t)(phase * THREADS + wid * WAVE) * 8),
+                16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][4] = {};
+    const int kg = (lane >> 4) * 8;
+    const int steps = kper / BK;
+
+    stage(0, 0);
+    if (steps > 1) stage(1, BK);
+    if (steps > 2) stage(2, 2 * BK);
+    for (int s = 0; s < steps; ++s) {
+        const int buf = s & 3;
+        if (s + 2 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(2 * S) : "memory");
+        else if (s + 1 < steps)
+            asm volatile("s_waitcnt vmcnt(%0)" ::"n"(S) : "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (s + 3 < steps) stage((s + 3) & 3, (s + 3) * BK);
+
+        const short* la = ldsA(buf);
+        const short* lb = ldsB(buf);
+#pragma unroll
+        for (int ks = 0; ks < BK / 32; ++ks) {
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                const int ar = wr + i * 16 + (lane & 15);
+                bf16x8 a = *(const bf16x8*)&la[ar * BK + ks * 32 + kg];
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const int bc = wc + j * 16 + (lane & 15);
+                    bf16x8 b = *(const bf16x8*)&lb[bc * BK + ks * 32 + kg];
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[i][j], 0, 0, 0);
+                }
+            }
+        }
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wr + i * 16 + (lane >> 4) * 4 + r;
+                int col = n0 + wc + j * 16 + (lane & 15);
+                if (row < M && col < N)
+                    unsafeAtomicAdd(&C[(size_t)row * N + col], acc[i][j][r]);
+            }
+        }
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(THREADS)
+gemm_bf16_128_splitk_kernel(const short* A, const short* Bt, float* C,
+                            int M, int N, int K, int ksplit) {
+    gemm_bf16_splitk_body<32>(A, Bt, C, M, N, K, ksplit);
+}
+
 extern "C" __global__ void __launch_bounds__(THREADS)
 gemm_bf16_128_pipe2_kernel(const short* A, const short* Bt, float* C, int M, int N, int K) {
     gemm_bf16_128_pipe2_body<32>(A, Bt, C, M, N, K);

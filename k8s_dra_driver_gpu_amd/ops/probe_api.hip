@@ -59,6 +59,7 @@ extern "C" __global__ void gemm_bf16_128_mfma32_bk64_kernel(const short*, const 
 extern "C" __global__ void gemm_bf16_128_pipe_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_128_pipe_bk64_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_128_pipe2_kernel(const short*, const short*, float*, int, int, int);
+extern "C" __global__ void gemm_bf16_128_splitk_kernel(const short*, const short*, float*, int, int, int, int);
 extern "C" __global__ void gemm_bf16_128_pipe3_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_128_pipe2r_kernel(const short*, const short*, float*, int, int, int);
 extern "C" __global__ void gemm_bf16_256_kernel(const short*, const short*, float*, int, int, int);
@@ -384,6 +385,72 @@ double fp_mfma_bf16_tflops(int dev, int inner_iters, int launches) {
 // ---------------------------------------------------------------------------
 // LDS-staged bf16 GEMM probe (gemm_probe.hip): throughput + host verify.
 // ---------------------------------------------------------------------------
+
+// Split-K bf16 GEMM for small shapes: ksplit partial products per output
+// tile, accumulated with f32 hardware atomics into a zeroed C. The memset
+// is part of each timed iteration (it is part of the op).
+double fp_gemm_bf16_splitk_tflops_mnk(int dev, int M, int N, int K,
+                                      int iters, int ksplit);
+
+double fp_gemm_bf16_splitk_tflops(int dev, int size, int iters, int ksplit) {
+    return fp_gemm_bf16_splitk_tflops_mnk(dev, size, size, size, iters, ksplit);
+}
+
+double fp_gemm_bf16_splitk_tflops_mnk(int dev, int M, int N, int K,
+                                      int iters, int ksplit) {
+    CHK(hipSetDevice(dev));
+    short *A, *Bt;
+    float* C;
+    CHK(hipMalloc(&A, (size_t)M * K * 2));
+    CHK(hipMalloc(&Bt, (size_t)N * K * 2));
+    CHK(hipMalloc(&C, (size_t)M * N * sizeof(float)));
+    CHK(hipMemset(A, 0x3f, (size_t)M * K * 2));
+    CHK(hipMemset(Bt, 0x3f, (size_t)N * K * 2));
+    dim3 grid((M / 128) * (N / 128) * ksplit);
+    hipEvent_t t0, t1;
+    CHK(hipEventCreate(&t0));
+    CHK(hipEventCreate(&t1));
+    CHK(hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float)));
+    hipLaunchKernelGGL(gemm_bf16_128_splitk_kernel, grid, dim3(256), 0, 0, A, Bt, C, M, N, K, ksplit);
+    CHK(hipGetLastError());
+    CHK(hipDeviceSynchronize());
+    CHK(hipEventRecord(t0));
+    for (int i = 0; i < iters; ++i) {
+        CHK(hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float)));
+        hipLaunchKernelGGL(gemm_bf16_128_splitk_kernel, grid, dim3(256), 0, 0, A, Bt, C, M, N, K, ksplit);
+    }
+    CHK(hipEventRecord(t1));
+    CHK(hipEventSynchronize(t1));
+    double ms = time_kernel_ms(t0, t1);
+    hipFree(A);
+    hipFree(Bt);
+    hipFree(C);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    return 2.0 * M * (double)N * K * iters / (ms * 1e9);
+}
+
+int fp_gemm_bf16_splitk_host(int dev, const short* A, const short* Bt,
+                             float* C, int M, int N, int K, int ksplit) {
+    CHKI(hipSetDevice(dev));
+    short *dA, *dB;
+    float* dC;
+    CHKI(hipMalloc(&dA, (size_t)M * K * 2));
+    CHKI(hipMalloc(&dB, (size_t)N * K * 2));
+    CHKI(hipMalloc(&dC, (size_t)M * N * sizeof(float)));
+    CHKI(hipMemcpy(dA, A, (size_t)M * K * 2, hipMemcpyHostToDevice));
+    CHKI(hipMemcpy(dB, Bt, (size_t)N * K * 2, hipMemcpyHostToDevice));
+    CHKI(hipMemset(dC, 0, (size_t)M * N * sizeof(float)));
+    dim3 grid(((M + 127) / 128) * ((N + 127) / 128) * ksplit);
+    hipLaunchKernelGGL(gemm_bf16_128_splitk_kernel, grid, dim3(256), 0, 0, dA, dB, dC, M, N, K, ksplit);
+    CHKI(hipGetLastError());
+    CHKI(hipDeviceSynchronize());
+    CHKI(hipMemcpy(C, dC, (size_t)M * N * sizeof(float), hipMemcpyDeviceToHost));
+    hipFree(dA);
+    hipFree(dB);
+    hipFree(dC);
+    return 0;
+}
 
 double fp_gemm_bf16_tflops_ex(int dev, int size, int iters, int bk) {
     CHK(hipSetDevice(dev));

@@ -458,6 +458,22 @@ class TestFp8Probe:
         assert tf8 > 1000, f"fp8 MX GEMM too slow: {tf8:.0f} TF"
         assert tf4 > 1500, f"fp4 MX GEMM too slow: {tf4:.0f} TF"
 
+    def test_splitk_numerics(self, probe):
+        # split-K partial products + f32 atomic accumulation vs the f64
+        # reference of bf16-truncated inputs; order-of-addition differences
+        # stay at fp32 rounding scale
+        rng = np.random.default_rng(7)
+        M = N = 512
+        K = 2048
+        a = rng.standard_normal((M, K)).astype(np.float32)
+        bt = rng.standard_normal((N, K)).astype(np.float32)
+        ref = (probe.bf16_truncate(a).astype(np.float64)
+               @ probe.bf16_truncate(bt).astype(np.float64).T)
+        for ks in (1, 4, 8):
+            c = probe.gemm_bf16_splitk(a, bt, ksplit=ks)
+            err = np.abs(c - ref).max() / np.abs(ref).max()
+            assert err < 1e-4, f"ksplit {ks}: {err}"
+
     def test_fp8_codec_round_trip(self, probe):
         # CPU-only property of the host codec, kept here with the fp8 suite
         rng = np.random.default_rng(0)
