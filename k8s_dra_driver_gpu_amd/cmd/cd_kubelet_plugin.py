@@ -18,6 +18,7 @@ from ..device.devicelib import DeviceLib
 from ..dra import api as dra
 from ..k8s.client import FakeClient, HttpClient
 from ..utils.debug import dump_config, install_stack_dump_handler
+from ..utils.paths import check_unix_socket_path
 
 logger = logging.getLogger("amddra.cmd.cd")
 
@@ -69,7 +70,7 @@ def main(argv=None) -> int:
         pass
     server = grpc.server(futures.ThreadPoolExecutor(max_workers=8))
     plugin.add_to_server(server)
-    server.add_insecure_port(f"unix://{dra_sock}")
+    server.add_insecure_port(f"unix://{check_unix_socket_path(dra_sock)}")
     server.start()
 
     registration = dra.RegistrationServicer(
@@ -83,7 +84,7 @@ def main(argv=None) -> int:
         pass
     reg_server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
     registration.add_to_server(reg_server)
-    reg_server.add_insecure_port(f"unix://{reg_sock}")
+    reg_server.add_insecure_port(f"unix://{check_unix_socket_path(reg_sock)}")
     reg_server.start()
 
     client.apply("resourceslices", plugin.resource_slice())

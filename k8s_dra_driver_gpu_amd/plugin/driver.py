@@ -130,7 +130,8 @@ class GpuDriver(dra.DRAPluginServicer):
         self._remove_stale(dra_sock)
         self._server = grpc.server(futures.ThreadPoolExecutor(max_workers=workers))
         self.add_to_server(self._server)
-        self._server.add_insecure_port(f"unix://{dra_sock}")
+        from ..utils.paths import check_unix_socket_path
+        self._server.add_insecure_port(f"unix://{check_unix_socket_path(dra_sock)}")
         self._server.start()
         out = {"dra": dra_sock}
 
@@ -143,7 +144,7 @@ class GpuDriver(dra.DRAPluginServicer):
             )
             self._reg_server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
             self.registration.add_to_server(self._reg_server)
-            self._reg_server.add_insecure_port(f"unix://{reg_sock}")
+            self._reg_server.add_insecure_port(f"unix://{check_unix_socket_path(reg_sock)}")
             self._reg_server.start()
             out["registration"] = reg_sock
         return out

@@ -72,11 +72,17 @@ class TestEntryPoints:
                 proc.kill()
 
     def test_cd_plugin_starts_and_stops(self, tmp_path):
+        import tempfile
+
         tree, env = _mock_env(tmp_path)
+        # socket dirs must stay short: AF_UNIX caps the path at 107 bytes
+        # and xdist's nested tmp_path exceeds it (the plugin now fails
+        # loudly for this; see utils/paths.check_unix_socket_path)
+        short = tempfile.mkdtemp(prefix="cdp-", dir="/tmp")
         env.update(
             {
-                "PLUGIN_DIR": str(tmp_path / "cdplugin"),
-                "PLUGINS_REGISTRY_DIR": str(tmp_path / "registry"),
+                "PLUGIN_DIR": os.path.join(short, "p"),
+                "PLUGINS_REGISTRY_DIR": os.path.join(short, "r"),
             }
         )
         proc = subprocess.Popen(
@@ -85,13 +91,15 @@ class TestEntryPoints:
         )
         try:
             deadline = time.monotonic() + 30
-            sock = tmp_path / "cdplugin" / "dra.sock"
+            from pathlib import Path
+            sock = Path(short) / "p" / "dra.sock"
             while time.monotonic() < deadline and not sock.exists():
                 assert proc.poll() is None, proc.stdout.read()
                 time.sleep(0.1)
             assert sock.exists()
             proc.send_signal(signal.SIGTERM)
-            assert proc.wait(timeout=10) == 0
+            rc = proc.wait(timeout=10)
+            assert rc == 0, f"exit {rc}: {proc.stdout.read()[-1500:]}"
         finally:
             if proc.poll() is None:
                 proc.kill()

@@ -351,3 +351,15 @@ class TestLogConfig:
         assert _logging.getLogger().level == _logging.DEBUG
         setup_logging(verbosity=4)
         assert _logging.getLogger().level == _logging.INFO
+
+
+class TestUnixSocketPathGuard:
+    def test_long_path_rejected_loudly(self):
+        import pytest as _pt
+
+        from k8s_dra_driver_gpu_amd.utils.paths import check_unix_socket_path
+
+        ok = "/var/lib/kubelet/plugins/gpu.amd.com/dra.sock"
+        assert check_unix_socket_path(ok) == ok
+        with _pt.raises(ValueError, match="AF_UNIX"):
+            check_unix_socket_path("/tmp/" + "x" * 110)
