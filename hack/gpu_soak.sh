@@ -30,11 +30,12 @@ cluster.client.create("computedomains", {
     "spec": {"numNodes": 1}})
 assert cluster.wait_cd_ready("soak-cd", "default", 60), "CD not ready"
 gpu = cluster.devicelib.gpus()[0]
-HOLD = int(os.environ.get("SOAK_HOLD", "0"))  # rolling window of HELD claims
-held = []  # (name, uid) of claims currently prepared (on distinct devices
-           # this would be real; on one GPU we hold PARTITION-free shared
-           # claims by using unique pod-level claims against the SAME gpu is
-           # double-booking — so held claims use the CD channel instead)
+# SOAK_HOLD: number of PrepareCompleted claims seeded into the live
+# checkpoint before churn starts. On a 1-GPU box we cannot hold real
+# prepared claims on distinct devices, so the hold models the actual
+# scaling concern: every churn RMW pays the cost of a large standing
+# checkpoint (see the fill() below).
+HOLD = int(os.environ.get("SOAK_HOLD", "0"))
 
 def churn_one(i):
     claim = cluster.client.create("resourceclaims", {
@@ -72,13 +73,6 @@ if HOLD:
         checkpoints=CheckpointManager(os.path.join(hd, "state")),
         state_dir=os.path.join(hd, "state"),
     )
-    # stack HOLD claims in one checkpoint file WITHOUT device overlap by
-    # using per-claim pseudo-devices? The overlap guard forbids same-GPU;
-    # hold the SAME claim re-prepared idempotently plus many completed
-    # entries via distinct claim uids against gpu-N partitions is blocked on
-    # ro sysfs. Simplest meaningful load: many claims in PrepareCompleted
-    # state written directly, so every subsequent RMW pays the real cost of
-    # a large checkpoint (the actual scaling concern).
     from k8s_dra_driver_gpu_amd.plugin.checkpoint import PreparedClaim, PreparedDevice, PREPARE_COMPLETED
     def fill(data):
         for i in range(HOLD):
