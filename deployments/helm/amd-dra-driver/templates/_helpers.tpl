@@ -59,3 +59,14 @@ resource.k8s.io/v1
 resource.k8s.io/v1beta1
 {{- end -}}
 {{- end }}
+
+{{/* Namespaces that run ComputeDomain daemon pods: the release namespace
+     plus controller.additionalNamespaces (comma-separated), mirroring the
+     controller's --additional-namespaces flag. */}}
+{{- define "amd-dra-driver.cdNamespaces" -}}
+{{- if .Values.controller.additionalNamespaces -}}
+{{- printf "%s,%s" (include "amd-dra-driver.namespace" .) .Values.controller.additionalNamespaces -}}
+{{- else -}}
+{{- include "amd-dra-driver.namespace" . -}}
+{{- end -}}
+{{- end }}

@@ -46,6 +46,10 @@ def daemon_set(
             "template": {
                 "metadata": {"labels": dict(labels)},
                 "spec": {
+                    # the chart creates this SA in every namespace the
+                    # controller may deploy daemons into
+                    # (rbac-compute-domain-daemon.yaml)
+                    "serviceAccountName": "amd-dra-cd-daemon",
                     # scheduled only onto nodes labeled into this CD
                     "nodeSelector": cd_label(cd_uid),
                     "containers": [

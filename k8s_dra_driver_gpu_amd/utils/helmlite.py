@@ -10,6 +10,7 @@ renders the same manifests a real ``helm template`` would:
 * ``{{ include "name" . }}`` (root context only),
 * ``.Values.*`` / ``.Release.Name`` / ``.Release.Namespace`` /
   ``.Chart.Name`` / ``.Chart.Version`` / ``.Chart.AppVersion`` lookups,
+* ``range $x := <list>`` / ``end`` over ``splitList`` results,
 * ``if`` / ``else`` / ``end`` with Helm truthiness (empty string, false,
   nil, 0 are falsy) and the ``eq``, ``ne``, ``and``, ``or``, ``not``
   functions over literals and lookups,
@@ -32,6 +33,9 @@ __all__ = ["render_chart", "render_template", "HelmliteError"]
 
 class HelmliteError(ValueError):
     pass
+
+
+_MISSING = object()
 
 
 _ACTION_RE = re.compile(r"\{\{-?\s*(.*?)\s*-?\}\}", re.DOTALL)
@@ -77,6 +81,7 @@ class _Renderer:
         self.chart = chart_meta
         self.release = release
         self.defines = defines
+        self.vars: Dict[str, Any] = {}  # $name bindings from range
 
     # -- expression evaluation ------------------------------------------
 
@@ -147,6 +152,11 @@ class _Renderer:
             return term == "true"
         if term.startswith(".") or term in (".", "$"):
             return self._lookup(term)
+        if term.startswith("$"):
+            name = term[1:]
+            if name not in self.vars:
+                raise HelmliteError(f"undefined variable {term!r}")
+            return self.vars[name]
         raise HelmliteError(f"unsupported term {term!r}")
 
     def _eval_expr(self, expr: str) -> Any:
@@ -187,6 +197,9 @@ class _Renderer:
             return v if _truthy(v) else self._eval_term(args[1])
         if head == "fail":
             raise HelmliteError(f"chart validation failed: {self._eval_term(args[1])}")
+        if head == "splitList":
+            sep = str(self._eval_term(args[1]))
+            return str(self._eval_term(args[2]) or "").split(sep)
         if head == "printf":
             fmt = self._eval_term(args[1])
             vals = tuple(self._eval_term(a) for a in args[2:])
@@ -278,6 +291,28 @@ class _Renderer:
                 rendered, _ = _Renderer._render_block(self, chosen, 0, None)
                 out.append(rendered)
                 continue
+            if word == "range":
+                # range $var := <pipeline>
+                m = re.fullmatch(r"range\s+\$(\w+)\s*:=\s*(.+)", payload, re.DOTALL)
+                if not m:
+                    raise HelmliteError(f"unsupported range form {payload!r}")
+                var, expr = m.group(1), m.group(2)
+                items = self._eval_pipeline(expr)
+                if items is None:
+                    items = []
+                if not isinstance(items, (list, tuple)):
+                    raise HelmliteError(f"range needs a list, got {type(items).__name__}")
+                body, i = self._collect_body(tokens, i + 1)
+                saved = self.vars.get(var, _MISSING)
+                for item in items:
+                    self.vars[var] = item
+                    rendered, _ = self._render_block(body, 0, None)
+                    out.append(rendered)
+                if saved is _MISSING:
+                    self.vars.pop(var, None)
+                else:
+                    self.vars[var] = saved
+                continue
             if word in ("end", "else"):
                 if until is None:
                     raise HelmliteError(f"unexpected {word!r}")
@@ -340,6 +375,23 @@ class _Renderer:
             cur.append(tokens[i])
             i += 1
         raise HelmliteError("unterminated if")
+
+    def _collect_body(self, tokens, i):
+        """Collect a range body up to its matching end."""
+        body: List = []
+        depth = 0
+        while i < len(tokens):
+            kind, payload = tokens[i]
+            word = payload.split(None, 1)[0] if kind == "action" and payload else ""
+            if kind == "action" and word in ("if", "define", "range", "with"):
+                depth += 1
+            elif kind == "action" and word == "end":
+                if depth == 0:
+                    return body, i + 1
+                depth -= 1
+            body.append(tokens[i])
+            i += 1
+        raise HelmliteError("unterminated range")
 
     def _skip_to_end(self, tokens, i):
         depth = 0

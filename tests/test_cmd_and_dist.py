@@ -362,6 +362,40 @@ class TestHelmChart:
                     f".Values.{r} not defined in values.yaml"
                 cur = cur[part]
 
+    def test_cd_daemon_rbac_and_serviceaccounts(self):
+        # the per-CD daemon pods need their own SA in EVERY namespace the
+        # controller may create DaemonSets in (release ns +
+        # controller.additionalNamespaces), bound to a ClusterRole covering
+        # cliques, CDs and pod labeling (ref rbac-compute-domain-daemon)
+        import yaml
+
+        docs = [d for d in yaml.safe_load_all(
+            self._render({"controller": {"additionalNamespaces": "t1,t2"}})[
+                "rbac-compute-domain-daemon.yaml"]) if d]
+        sas = [d for d in docs if d["kind"] == "ServiceAccount"]
+        assert {d["metadata"]["namespace"] for d in sas} == {
+            "amd-dra-driver", "t1", "t2"}
+        assert all(d["metadata"]["name"] == "amd-dra-cd-daemon" for d in sas)
+        role = [d for d in docs if d["kind"] == "ClusterRole"][0]
+        got = {(g, r): set(rule["verbs"]) for rule in role["rules"]
+               for g in rule["apiGroups"] for r in rule["resources"]}
+        assert "create" in got[("resource.amd.com", "computedomaincliques")]
+        assert "patch" in got[("", "pods")]
+        assert "update" in got[("resource.amd.com", "computedomains/status")]
+        crb = [d for d in docs if d["kind"] == "ClusterRoleBinding"][0]
+        assert {s["namespace"] for s in crb["subjects"]} == {
+            "amd-dra-driver", "t1", "t2"}
+        # the runtime-rendered DaemonSet runs as exactly that SA
+        from k8s_dra_driver_gpu_amd.controller.templates import daemon_set
+
+        ds = daemon_set("cd1", "uid1", "default")
+        assert (ds["spec"]["template"]["spec"]["serviceAccountName"]
+                == "amd-dra-cd-daemon")
+        # default (no additional namespaces): single SA in the release ns
+        docs1 = [d for d in yaml.safe_load_all(
+            self._render()["rbac-compute-domain-daemon.yaml"]) if d]
+        assert len([d for d in docs1 if d["kind"] == "ServiceAccount"]) == 1
+
     def test_rbac_covers_driver_resources(self):
         docs = self._docs()
         rules = []

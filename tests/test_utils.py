@@ -363,3 +363,32 @@ class TestUnixSocketPathGuard:
         assert check_unix_socket_path(ok) == ok
         with _pt.raises(ValueError, match="AF_UNIX"):
             check_unix_socket_path("/tmp/" + "x" * 110)
+
+
+class TestHelmliteRange:
+    def test_range_splitlist(self):
+        from k8s_dra_driver_gpu_amd.utils.helmlite import render_template
+
+        out = render_template(
+            '{{- range $n := splitList "," "a,b" }}\nx: {{ $n | quote }}\n{{- end }}', {})
+        assert out == '\nx: "a"\nx: "b"'
+
+    def test_range_variable_scoping_restores(self):
+        from k8s_dra_driver_gpu_amd.utils.helmlite import (
+            HelmliteError, render_template,
+        )
+        import pytest as _pt
+
+        # $n is undefined outside the range body
+        with _pt.raises(HelmliteError, match="undefined variable"):
+            render_template(
+                '{{- range $n := splitList "," "a" }}{{ $n }}{{- end }}{{ $n }}', {})
+
+    def test_nested_if_inside_range(self):
+        from k8s_dra_driver_gpu_amd.utils.helmlite import render_template
+
+        out = render_template(
+            '{{- range $n := splitList "," "a,b" }}'
+            '{{- if eq $n "a" }}[{{ $n }}]{{- else }}({{ $n }}){{- end }}'
+            '{{- end }}', {})
+        assert out == "[a](b)"
