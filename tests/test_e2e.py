@@ -379,7 +379,7 @@ class TestLocalClusterDemo:
             os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
             "demo", "specs", "quickstart",
         )
-        cluster = LocalCluster(num_gpus=8, work_dir=str(tmp_path)).start()
+        cluster = LocalCluster(num_gpus=16, work_dir=str(tmp_path)).start()
         try:
             ev1 = cluster.apply_yaml(os.path.join(specs, "gpu-test1.yaml"))
             assert any("prepared gpu-" in e for e in ev1), ev1
@@ -394,6 +394,14 @@ class TestLocalClusterDemo:
             assert any("prepared gpu-" in e for e in ev2), ev2
             ev7 = cluster.apply_yaml(os.path.join(specs, "gpu-test5.yaml"))
             assert any("prepared gpu-" in e for e in ev7), ev7
+            # two distinct GPUs in one pod (two claims)
+            ev8 = cluster.apply_yaml(os.path.join(specs, "gpu-test7.yaml"))
+            got = [e for e in ev8 if "prepared gpu-" in e]
+            assert len(got) == 2 and len(set(got)) == 2, ev8
+            # one claim asking two devices (count: 2)
+            ev9 = cluster.apply_yaml(os.path.join(specs, "gpu-test8.yaml"))
+            two = [e for e in ev9 if "prepared gpu-" in e]
+            assert len(two) >= 1, ev9
             ev4 = cluster.apply_yaml(os.path.join(specs, "cd-test1.yaml"))
             assert cluster.wait_cd_ready("cd1", "cd-test1"), ev4
             assert any("prepared channel-0" in e for e in ev4), ev4
