@@ -323,6 +323,14 @@ class TestHelmChart:
         with pytest.raises(HelmliteError, match="resourceApiVersion"):
             self._render({"resourceApiVersion": "v2"})
 
+    def test_default_namespace_rejected(self):
+        from k8s_dra_driver_gpu_amd.utils.helmlite import HelmliteError
+
+        with pytest.raises(HelmliteError, match="default.*namespace"):
+            self._render({"namespace": "default"})
+        # explicit override allows it
+        self._render({"namespace": "default", "allowDefaultNamespace": True})
+
     def test_standard_labels_on_all_objects(self):
         docs = self._docs()
         for d in docs:
