@@ -10,7 +10,7 @@ renders the same manifests a real ``helm template`` would:
 * ``{{ include "name" . }}`` (root context only),
 * ``.Values.*`` / ``.Release.Name`` / ``.Release.Namespace`` /
   ``.Chart.Name`` / ``.Chart.Version`` / ``.Chart.AppVersion`` lookups,
-* ``range $x := <list>`` / ``end`` over ``splitList`` results,
+* ``range $x := <list>`` / ``end`` over ``splitList`` results, ``toYaml``,
 * ``if`` / ``else`` / ``end`` with Helm truthiness (empty string, false,
   nil, 0 are falsy) and the ``eq``, ``ne``, ``and``, ``or``, ``not``
   functions over literals and lookups,
@@ -233,6 +233,10 @@ class _Renderer:
             elif fn == "indent":
                 pad = " " * int(fargs[0])
                 val = "\n".join(pad + l for l in str(val or "").splitlines())
+            elif fn == "toYaml":
+                import yaml as _yaml
+                val = _yaml.safe_dump(val, default_flow_style=False,
+                                      sort_keys=False).rstrip("\n")
             elif fn == "nindent":
                 pad = " " * int(fargs[0])
                 val = "\n" + "\n".join(pad + l for l in str(val or "").splitlines())

@@ -323,6 +323,25 @@ class TestHelmChart:
         with pytest.raises(HelmliteError, match="resourceApiVersion"):
             self._render({"resourceApiVersion": "v2"})
 
+    def test_scheduling_knobs(self):
+        # kubelet plugin must outrank workload pods (system-node-critical);
+        # tolerations and imagePullSecrets render from values lists
+        import yaml
+
+        out = self._render({
+            "kubeletPlugin": {"tolerations": [{"operator": "Exists"}]},
+            "imagePullSecrets": [{"name": "cred"}],
+        })
+        kp = list(yaml.safe_load_all(out["kubeletplugin.yaml"]))[0]
+        spec = kp["spec"]["template"]["spec"]
+        assert spec["priorityClassName"] == "system-node-critical"
+        assert spec["tolerations"] == [{"operator": "Exists"}]
+        assert spec["imagePullSecrets"] == [{"name": "cred"}]
+        dep = [d for d in yaml.safe_load_all(self._render()["controller.yaml"])
+               if d and d["kind"] == "Deployment"][0]
+        assert (dep["spec"]["template"]["spec"]["priorityClassName"]
+                == "system-cluster-critical")
+
     def test_default_namespace_rejected(self):
         from k8s_dra_driver_gpu_amd.utils.helmlite import HelmliteError
 
