@@ -592,6 +592,67 @@ class TestCdPluginTwoPhase:
         assert plugin.cleanup_stale_claims() >= 1
         assert plugin.checkpoints.load().get_claim(UID_CLAIM) is None
 
+    def test_randomized_crash_injection_recovers(self, cd_plugin):
+        """Seeded fuzz over the 2-phase protocol: fail the CDI spec write or
+        the domain-dir creation on a random call, verify the claim errors
+        without corrupting state, then retry cleanly, unprepare, and assert
+        nothing is left behind (no claim spec, no checkpoint entry)."""
+        import glob
+        import random
+
+        client, lib, plugin = cd_plugin
+        rng = random.Random(4242)
+        for trial in range(12):
+            cd = make_cd(client, name=f"cdz{trial}")
+            uid = cd["metadata"]["uid"]
+            claim_uid = f"{trial:08d}-fuzz-4000-8000-000000000000"
+            kind = rng.choice(["daemon", "channel"])
+            if kind == "channel":
+                self._ready_clique(client, plugin, uid)
+                _mk_claim(client, "default", f"cz{trial}", claim_uid,
+                          "channel-0", "ComputeDomainChannelConfig", uid)
+            else:
+                _mk_claim(client, "default", f"cz{trial}", claim_uid,
+                          "daemon-0", "ComputeDomainDaemonConfig", uid)
+            req = dra.NodePrepareResourcesRequest(
+                claims=[dra.Claim(namespace="default", name=f"cz{trial}",
+                                  uid=claim_uid)])
+
+            target = rng.choice(["cdi", "dir"])
+            real_write = plugin.cdi.write_claim_spec
+            real_dir = plugin._ensure_domain_dir
+
+            def boom(*a, **k):
+                raise OSError("injected crash")
+
+            if target == "cdi":
+                plugin.cdi.write_claim_spec = boom
+            else:
+                plugin._ensure_domain_dir = boom
+            try:
+                resp = plugin.node_prepare_resources(req, None)
+                assert resp.claims[claim_uid].error != ""
+            finally:
+                plugin.cdi.write_claim_spec = real_write
+                plugin._ensure_domain_dir = real_dir
+
+            # retry with the fault cleared must complete
+            resp = plugin.node_prepare_resources(req, None)
+            assert resp.claims[claim_uid].error == "", resp.claims[claim_uid].error
+            cp = plugin.checkpoints.load()
+            assert cp.get_claim(claim_uid).state == "PrepareCompleted"
+
+            # unprepare leaves no residue for this claim
+            plugin.node_unprepare_resources(
+                dra.NodeUnprepareResourcesRequest(
+                    claims=[dra.Claim(namespace="default", name=f"cz{trial}",
+                                      uid=claim_uid)]), None)
+            assert plugin.checkpoints.load().get_claim(claim_uid) is None
+            leftovers = [f for f in glob.glob(
+                os.path.join(plugin.cdi.cdi_root, "*.json"))
+                if claim_uid in f]
+            assert not leftovers, leftovers
+
     def test_unprepare_of_started_claim_is_clean(self, cd_plugin):
         client, lib, plugin = cd_plugin
         from k8s_dra_driver_gpu_amd.plugin.checkpoint import ClaimRef, PreparedClaim
