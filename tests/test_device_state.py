@@ -251,6 +251,53 @@ class TestCrashRecovery:
         assert len(r) == 1
         assert lib.gpu_by_minor(0).compute_partition == SPX
 
+    def test_randomized_crash_injection_recovers(self, env):
+        """Seeded fuzz: fail the CDI claim-spec write or the partition write
+        at a random prepare, verify the claim fails atomically (no
+        checkpoint residue, device back in SPX), then a clean retry
+        succeeds and unprepare leaves nothing behind."""
+        import random
+
+        tree, lib, cdi, cps, ds = env
+        rng = random.Random(777)
+        for trial in range(10):
+            uid = f"{trial:08d}-dsfz-4000-8000-000000000000"
+            dev = rng.choice(["gpu-0", "gpu-0-cpx-3", "gpu-8"])
+            target = rng.choice(["cdi", "sysfs"])
+            real_write = cdi.write_claim_spec
+            real_set = lib.backend.set_compute_partition
+
+            def boom(*a, **k):
+                raise OSError("injected crash")
+
+            if target == "cdi":
+                cdi.write_claim_spec = boom
+            elif "cpx" in dev:
+                lib.backend.set_compute_partition = boom
+            try:
+                if target == "cdi" or "cpx" in dev:
+                    with pytest.raises(Exception):
+                        ds.prepare(claim(uid, dev))
+                else:
+                    ds.prepare(claim(uid, dev))
+                    ds.unprepare(uid)
+                    continue
+            finally:
+                cdi.write_claim_spec = real_write
+                lib.backend.set_compute_partition = real_set
+            # atomic failure: no completed checkpoint entry survives
+            pc = cps.load().get_claim(uid)
+            assert pc is None or pc.state == PREPARE_STARTED
+            lib.invalidate()
+            # clean retry
+            r = ds.prepare(claim(uid, dev))
+            assert len(r) == 1
+            ds.unprepare(uid)
+            assert cps.load().get_claim(uid) is None
+            assert not os.path.exists(cdi.claim_spec_path(uid))
+            lib.invalidate()
+            assert lib.gpu_by_minor(0).compute_partition == SPX
+
     def test_destroy_unknown_partitions(self, env):
         tree, lib, _, _, ds = env
         lib.backend.set_compute_partition(8, CPX)  # second GPU = card minor 8
