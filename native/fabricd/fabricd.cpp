@@ -474,14 +474,20 @@ static void run_gpu_probe() {
         return;
     }
     double gbps = hbm(0, (size_t)1 << 30, 3);
+    // optional: MX quantized-GEMM floor (real per-block scales through the
+    // matrix cores) — present in current probe builds, tolerated absent
+    auto mx = (double (*)(int, int, int, int))dlsym(h, "fp_gemm_fp8_scaled_tflops");
+    double mxtf = mx ? mx(0, 2048, 3, 556) : 0.0;
+    char mxs[64] = "";
+    if (mxtf > 0) snprintf(mxs, sizeof mxs, " gemm_fp8_mx=%.0fTF", mxtf);
     char buf[256];
     if (n >= 2) {
         double agbps = ar((size_t)512 << 20, 3);
-        snprintf(buf, sizeof buf, "gpus=%d hbm_read=%.0fGB/s allreduce_pull=%.0fGB/s", n,
-                 gbps, agbps);
+        snprintf(buf, sizeof buf, "gpus=%d hbm_read=%.0fGB/s allreduce_pull=%.0fGB/s%s", n,
+                 gbps, agbps, mxs);
         g_probe_ok = gbps > 100 && agbps > 10;
     } else {
-        snprintf(buf, sizeof buf, "gpus=%d hbm_read=%.0fGB/s", n, gbps);
+        snprintf(buf, sizeof buf, "gpus=%d hbm_read=%.0fGB/s%s", n, gbps, mxs);
         g_probe_ok = gbps > 100;
     }
     g_probe_report = buf;
