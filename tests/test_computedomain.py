@@ -797,19 +797,32 @@ class TestFabricd:
         )
         return out.stdout.strip()
 
-    def test_two_daemon_mesh_over_hostnames(self, tmp_path):
-        """FabricDaemonsWithDNSNames data path: peers given as RESOLVABLE
-        HOSTNAMES (localhost standing in for the headless-service DNS names)
-        — exercises fabricd's getaddrinfo path, not just numeric IPs."""
-        d1, d2 = str(tmp_path / "a"), str(tmp_path / "b")
+    def _launch_pair(self, tmp_path, host, attempt=0):
+        """Start a 2-daemon mesh; retry once with fresh ports if either
+        daemon dies at startup (a _free_port can be re-grabbed between the
+        probe-close and fabricd's bind)."""
+        d1, d2 = str(tmp_path / f"a{attempt}"), str(tmp_path / f"b{attempt}")
         os.makedirs(d1), os.makedirs(d2)
         p1, c1, p2, c2 = _free_port(), _free_port(), _free_port(), _free_port()
-        self._write_cfg(d1, p1, c1, [f"localhost:{p2}"])
-        self._write_cfg(d2, p2, c2, [f"localhost:{p1}"])
+        self._write_cfg(d1, p1, c1, [f"{host}:{p2}"])
+        self._write_cfg(d2, p2, c2, [f"{host}:{p1}"])
         procs = [
             subprocess.Popen([FABRICD, "-c", os.path.join(d, "fabricd.cfg")])
             for d in (d1, d2)
         ]
+        time.sleep(0.5)
+        if any(p.poll() is not None for p in procs) and attempt == 0:
+            for p in procs:
+                p.kill()
+                p.wait(timeout=5)
+            return self._launch_pair(tmp_path, host, attempt=1)
+        return procs, c1, c2
+
+    def test_two_daemon_mesh_over_hostnames(self, tmp_path):
+        """FabricDaemonsWithDNSNames data path: peers given as RESOLVABLE
+        HOSTNAMES (localhost standing in for the headless-service DNS names)
+        — exercises fabricd's getaddrinfo path, not just numeric IPs."""
+        procs, c1, c2 = self._launch_pair(tmp_path, "localhost")
         try:
             ok = wait_for(
                 lambda: self._status(c1).startswith("READY") and self._status(c2).startswith("READY"),
@@ -823,15 +836,7 @@ class TestFabricd:
                 p.wait(timeout=5)
 
     def test_two_daemon_mesh_reaches_ready(self, tmp_path):
-        d1, d2 = str(tmp_path / "a"), str(tmp_path / "b")
-        os.makedirs(d1), os.makedirs(d2)
-        p1, c1, p2, c2 = _free_port(), _free_port(), _free_port(), _free_port()
-        self._write_cfg(d1, p1, c1, [f"127.0.0.1:{p2}"])
-        self._write_cfg(d2, p2, c2, [f"127.0.0.1:{p1}"])
-        procs = [
-            subprocess.Popen([FABRICD, "-c", os.path.join(d, "fabricd.cfg")])
-            for d in (d1, d2)
-        ]
+        procs, c1, c2 = self._launch_pair(tmp_path, "127.0.0.1")
         try:
             ok = wait_for(
                 lambda: self._status(c1).startswith("READY") and self._status(c2).startswith("READY"),
