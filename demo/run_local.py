@@ -17,10 +17,11 @@ SPECS = os.path.join(os.path.dirname(os.path.abspath(__file__)), "specs", "quick
 
 def main() -> int:
     logging.basicConfig(level=logging.WARNING)
-    cluster = LocalCluster(num_gpus=8, vfio=True).start()
+    cluster = LocalCluster(num_gpus=16, vfio=True).start()
     try:
         for spec in ("gpu-test1.yaml", "gpu-test2.yaml", "gpu-test3.yaml",
-                     "gpu-test5.yaml", "gpu-test-partitions.yaml",
+                     "gpu-test5.yaml", "gpu-test7.yaml", "gpu-test8.yaml",
+                     "gpu-test-partitions.yaml",
                      "gpu-test-vfio.yaml", "gpu-test-extres.yaml"):
             print(f"=== {spec} ===")
             for ev in cluster.apply_yaml(os.path.join(SPECS, spec)):
